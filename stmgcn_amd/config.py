@@ -1,0 +1,92 @@
+"""Configuration for the ST-MGCN MI355X stack.
+
+The reference keeps hyperparameters as module constants (Main.py:9-17) and
+architecture literals inline (n_nodes=58 at Main.py:62). Here they are lifted
+into one dataclass; PRESETS carries the five BASELINE.json configs.
+"""
+from __future__ import annotations
+
+import dataclasses
+from dataclasses import dataclass, field
+from typing import List, Optional
+
+
+@dataclass
+class STMGCNConfig:
+    # --- model architecture (reference Main.py:62, STMGCN.py:61-78) ---
+    n_nodes: int = 58
+    input_dim: int = 1
+    seq_len: int = 5                 # = sum(obs_len); reference default obs (3,1,1)
+    lstm_hidden_dim: int = 64
+    lstm_num_layers: int = 3
+    gcn_hidden_dim: int = 64
+    m_graphs: int = 3                # M static region graphs
+    kernel_type: str = "chebyshev"   # chebyshev | localpool | random_walk_diffusion
+    cheby_K: int = 2
+    rnn_cell: str = "lstm"           # lstm | gru  (gru for the deep variant)
+    n_blocks: int = 1                # stacked ST-MGCN blocks (deep variant = 4)
+    gconv_use_bias: bool = True
+    gconv_activation: str = "relu"
+
+    # --- numerics ---
+    dtype: str = "fp32"              # fp32 | bf16 | fp16 (compute dtype on GPU)
+    # lambda_max handling for Chebyshev rescaling: the reference's torch.eig is
+    # dead on modern torch so lambda_max == 2 ALWAYS (GCN.py:117-121, SURVEY
+    # quirk 1). "fixed2" replicates that; "power_iteration" computes it.
+    lambda_max_mode: str = "fixed2"
+
+    # --- training (reference Main.py:9-17) ---
+    batch_size: int = 32             # per-rank batch
+    lr: float = 2e-3
+    weight_decay: float = 1e-4
+    n_epochs: int = 100
+    loss: str = "MSE"                # MSE | MAE | Huber
+    early_stop_patience: int = 10
+    shuffle: bool = False            # reference never shuffles (quirk 5)
+
+    # --- data windowing (reference Main.py:30-33, Data_Container.py:125-146) ---
+    obs_len: List[int] = field(default_factory=lambda: [3, 1, 1])  # serial, daily, weekly
+    dt: int = 1                      # hours per timestep
+
+    # --- parallelism ---
+    dp_degree: int = 1
+
+    @property
+    def support_K(self) -> int:
+        """Support count per graph — the reference contract (STMGCN.py:80-91):
+        localpool -> 1, chebyshev -> K+1, random_walk_diffusion -> 2K+1.
+        NOTE: the rw-diffusion preprocessor actually emits K+1 supports
+        (GCN.py:77-81), so that kernel type is unusable end-to-end in the
+        reference; we replicate the contract and document it.
+        """
+        if self.kernel_type == "localpool":
+            return 1
+        if self.kernel_type == "chebyshev":
+            return self.cheby_K + 1
+        if self.kernel_type == "random_walk_diffusion":
+            return 2 * self.cheby_K + 1
+        raise ValueError(f"unknown kernel_type {self.kernel_type!r}")
+
+    def replace(self, **kw) -> "STMGCNConfig":
+        return dataclasses.replace(self, **kw)
+
+
+# The five BASELINE.json configs.
+PRESETS = {
+    # config[0]: plumbing, CPU
+    "cpu-small": STMGCNConfig(n_nodes=64, seq_len=5, cheby_K=2, dtype="fp32",
+                              m_graphs=1, batch_size=8),
+    # reference defaults (58 regions, for parity work)
+    "reference": STMGCNConfig(),
+    # config[1]/[2]: the headline bench config — 3-graph, 1024 regions, seq 8
+    "bench-1024": STMGCNConfig(n_nodes=1024, seq_len=8, cheby_K=2, dtype="bf16",
+                               m_graphs=3, batch_size=32,
+                               obs_len=[4, 2, 2]),
+    # config[3]: deep variant — 4 stacked blocks, GRU CGRNN, 4096 regions, fp16
+    "deep-4096": STMGCNConfig(n_nodes=4096, seq_len=8, cheby_K=2, dtype="fp16",
+                              m_graphs=3, batch_size=16, rnn_cell="gru",
+                              n_blocks=4, obs_len=[4, 2, 2]),
+    # config[4]: large-city — 16384 regions, K=3, batch packed to HBM
+    "large-16384": STMGCNConfig(n_nodes=16384, seq_len=8, cheby_K=3, dtype="bf16",
+                                m_graphs=3, batch_size=64, obs_len=[4, 2, 2]),
+}

@@ -1,0 +1,2 @@
+from .trainer import ModelTrainer  # noqa: F401
+from .metrics import MSE, RMSE, MAE, MAPE, PCC  # noqa: F401

@@ -1,0 +1,179 @@
+"""Training runtime — API and observable-behavior parity with the reference
+ModelTrainer (Model_Trainer.py:8-114), rebuilt for MI355X.
+
+Parity kept:
+  - constructor signature (model, loss, optimizer, lr, wd, n_epochs)
+    (Model_Trainer.py:9-14; `optimizer` is the class, e.g. torch.optim.Adam)
+  - train(data_loader, sta_adj_list, modes, model_dir, early_stopper=10):
+    best-val checkpointing to {model_dir}/{cls}_best_model.pkl with payload
+    {'epoch': int, 'state_dict': ...}; patience reset on improvement;
+    exact epoch print strings (Model_Trainer.py:47-60)
+  - test(...) reloads the best checkpoint, prints denormalized
+    MSE/RMSE/MAE/MAPE with the reference's formats (Model_Trainer.py:92-95)
+  - class-name dispatch: only 'ST_MGCN'-named models accepted (quirk 9)
+
+Reference quirks intentionally FIXED (documented in SURVEY Appendix A):
+  - running_loss accumulates a detached float, not a live tensor (quirk 6)
+  - test()/validate run under torch.no_grad() (quirk 7)
+
+MI355X additions:
+  - data-parallel awareness: an optional GradReducer all-reduces gradients
+    after backward; loss scalars are all-reduced for identical early-stop
+    decisions on every rank; only rank 0 writes checkpoints/prints
+  - per-step HIP-event timing and JSONL metrics stream (samples/sec, val
+    loss) for the scaling-curve report (SURVEY §5 observability)
+"""
+from __future__ import annotations
+
+import json
+import os
+import time
+from typing import Dict, List, Optional
+
+import numpy as np
+import torch
+from torch import nn
+
+from .metrics import MSE, RMSE, MAE, MAPE, PCC
+
+
+class ModelTrainer:
+    SUPPORTED_MODELS = ("ST_MGCN", "StackedSTMGCN")
+
+    def __init__(self, model: nn.Module, loss: nn.Module, optimizer, lr: float,
+                 wd: float, n_epochs: int, grad_reducer=None, rank: int = 0,
+                 world_size: int = 1, metrics_path: Optional[str] = None):
+        self.model = model
+        self.model_name = model.__class__.__name__
+        if self.model_name not in self.SUPPORTED_MODELS:
+            raise ValueError(f"unsupported model class {self.model_name!r}")
+        self.criterion = loss
+        self.optimizer = optimizer(params=model.parameters(), lr=lr, weight_decay=wd)
+        self.n_epochs = n_epochs
+        self.grad_reducer = grad_reducer
+        self.rank, self.world = rank, world_size
+        self.metrics_path = metrics_path
+        self._metrics_f = None
+
+    # ------------------------------------------------------------------ utils
+    def _log(self, *args):
+        if self.rank == 0:
+            print(*args)
+
+    def _emit(self, record: dict):
+        if self.rank != 0 or self.metrics_path is None:
+            return
+        if self._metrics_f is None:
+            os.makedirs(os.path.dirname(self.metrics_path) or ".", exist_ok=True)
+            self._metrics_f = open(self.metrics_path, "a")
+        self._metrics_f.write(json.dumps(record) + "\n")
+        self._metrics_f.flush()
+
+    def _allreduce_scalar(self, value: float) -> float:
+        if self.world <= 1:
+            return value
+        import torch.distributed as dist
+        t = torch.tensor([value], dtype=torch.float64)
+        if torch.distributed.is_initialized():
+            dist.all_reduce(t)
+            return float(t.item()) / self.world
+        return value
+
+    def _forward(self, x, sta_adj_list):
+        return self.model(obs_seq=x, sta_adj_list=sta_adj_list)
+
+    def _ckpt_path(self, model_dir: str) -> str:
+        return os.path.join(model_dir, f"{self.model_name}_best_model.pkl")
+
+    # ------------------------------------------------------------------ train
+    def train(self, data_loader: Dict, sta_adj_list: List, modes: List[str],
+              model_dir: str, early_stopper: int = 10):
+        patience = early_stopper
+        checkpoint = {"epoch": 0, "state_dict": self.model.state_dict()}
+        val_loss = np.inf
+        self._log("Training starts at: ", time.ctime())
+
+        for epoch in range(1, self.n_epochs + 1):
+            running_loss = {mode: 0.0 for mode in modes}
+            t_epoch = time.perf_counter()
+            n_train_samples = 0
+            for mode in modes:
+                self.model.train() if mode == "train" else self.model.eval()
+                step = 0
+                for x, y_true in data_loader[mode]:
+                    with torch.set_grad_enabled(mode == "train"):
+                        y_pred = self._forward(x, sta_adj_list)
+                        loss = self.criterion(y_pred, y_true)
+                        if mode == "train":
+                            if self.grad_reducer is not None:
+                                self.grad_reducer.zero_grad()
+                            else:
+                                self.optimizer.zero_grad(set_to_none=True)
+                            loss.backward()
+                            if self.grad_reducer is not None:
+                                self.grad_reducer.reduce()
+                            self.optimizer.step()
+                    running_loss[mode] += float(loss.detach()) * y_true.shape[0]
+                    step += y_true.shape[0]
+                if mode == "train":
+                    n_train_samples = step * self.world
+
+                if mode == "validate":
+                    epoch_val = self._allreduce_scalar(running_loss[mode] / max(step, 1))
+                    dt = time.perf_counter() - t_epoch
+                    self._emit({"epoch": epoch, "val_loss": epoch_val,
+                                "train_loss": running_loss.get("train", 0.0) / max(n_train_samples // max(self.world, 1), 1),
+                                "samples_per_sec": n_train_samples / dt if dt > 0 else 0.0,
+                                "wall_s": dt})
+                    if epoch_val <= val_loss:
+                        self._log(f"Epoch {epoch}, Val_loss drops from {val_loss:.5} "
+                                  f"to {epoch_val:.5}. Update model checkpoint..")
+                        val_loss = epoch_val
+                        checkpoint.update(epoch=epoch, state_dict=self.model.state_dict())
+                        if self.rank == 0:
+                            torch.save(checkpoint, self._ckpt_path(model_dir))
+                        patience = early_stopper
+                    else:
+                        self._log(f"Epoch {epoch}, Val_loss does not improve from {val_loss:.5}.")
+                        patience -= 1
+                        if patience == 0:
+                            self._log(f"Early stopping at epoch {epoch}..")
+                            return
+
+        self._log("Training ends at: ", time.ctime())
+        if self.rank == 0:
+            torch.save(checkpoint, self._ckpt_path(model_dir))
+
+    # ------------------------------------------------------------------- test
+    def test(self, data_loader: Dict, sta_adj_list: List, modes: List[str],
+             model_dir: str, data_class):
+        saved = torch.load(self._ckpt_path(model_dir), weights_only=False)
+        self.model.load_state_dict(saved["state_dict"])
+        self.model.eval()
+        self._log("Testing starts at: ", time.ctime())
+        results = {}
+        with torch.no_grad():
+            for mode in modes:
+                truths, preds = [], []
+                for x, y_true in data_loader[mode]:
+                    y_pred = self._forward(x, sta_adj_list)
+                    truths.append(y_true.float().cpu().numpy())
+                    preds.append(y_pred.float().cpu().numpy())
+                gt = data_class.minmax_denormalize(np.concatenate(truths, axis=0))
+                pr = data_class.minmax_denormalize(np.concatenate(preds, axis=0))
+                results[mode] = {"MSE": MSE(pr, gt), "RMSE": RMSE(pr, gt),
+                                 "MAE": MAE(pr, gt), "MAPE": MAPE(pr, gt)}
+                self._log(f"{mode} true MSE: ", results[mode]["MSE"])
+                self._log(f"{mode} true RMSE: ", results[mode]["RMSE"])
+                self._log(f"{mode} true MAE: ", results[mode]["MAE"])
+                self._log(f"{mode} true MAPE: ", results[mode]["MAPE"] * 100, "%")
+                self._emit({"test_mode": mode, **results[mode]})
+        self._log("Testing ends at: ", time.ctime())
+        return results
+
+    # static metric surface (reference Model_Trainer.py:100-114)
+    MSE = staticmethod(MSE)
+    RMSE = staticmethod(RMSE)
+    MAE = staticmethod(MAE)
+    MAPE = staticmethod(MAPE)
+    PCC = staticmethod(PCC)

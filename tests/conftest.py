@@ -1,0 +1,20 @@
+import os
+import sys
+
+import pytest
+import torch
+
+sys.path.insert(0, os.path.dirname(os.path.dirname(os.path.abspath(__file__))))
+
+
+def pytest_configure(config):
+    config.addinivalue_line(
+        "markers", "gpu: test requires an MI355X GPU (run via gpurun / driver)")
+
+
+def pytest_collection_modifyitems(config, items):
+    if not torch.cuda.is_available():
+        skip_gpu = pytest.mark.skip(reason="no GPU in this environment")
+        for item in items:
+            if "gpu" in item.keywords:
+                item.add_marker(skip_gpu)

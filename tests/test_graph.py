@@ -1,0 +1,102 @@
+"""Graph preprocessing tests: support stacks, CSR
+generator round-trip, support-count contract, reference-quirk behavior."""
+import numpy as np
+import pytest
+import torch
+
+from stmgcn_amd.graph import SupportGenerator, CSRSupport
+from stmgcn_amd.graph.preprocess import (
+    symmetric_normalize, random_walk_normalize, chebyshev_polynomials,
+    dense_to_csr, power_iteration_lmax)
+from stmgcn_amd.data.synthetic import _random_sparse_sym_adj
+
+
+def _adj(n=32, seed=0):
+    rng = np.random.default_rng(seed)
+    return torch.from_numpy(_random_sparse_sym_adj(n, 6, rng, weighted=True))
+
+
+def test_symmetric_normalize_matches_definition():
+    A = _adj()
+    D = torch.diag(A.sum(1).pow(-0.5))
+    expected = D @ A @ D
+    torch.testing.assert_close(symmetric_normalize(A), expected, rtol=1e-6, atol=1e-6)
+
+
+def test_symmetric_normalize_isolated_node_guard():
+    A = torch.zeros(4, 4)
+    A[0, 1] = A[1, 0] = 1.0  # nodes 2,3 isolated
+    out = symmetric_normalize(A)
+    assert torch.isfinite(out).all()
+
+
+def test_random_walk_normalize_rows_sum_to_one():
+    A = _adj()
+    P = random_walk_normalize(A)
+    torch.testing.assert_close(P.sum(1), torch.ones(A.shape[0]), rtol=1e-5, atol=1e-5)
+
+
+def test_chebyshev_recurrence():
+    G = symmetric_normalize(_adj())
+    polys = chebyshev_polynomials(G, 3)
+    N = G.shape[0]
+    torch.testing.assert_close(polys[0], torch.eye(N))
+    torch.testing.assert_close(polys[1], G)
+    torch.testing.assert_close(polys[2], 2 * G @ G - torch.eye(N), rtol=1e-5, atol=1e-5)
+    torch.testing.assert_close(polys[3], 2 * G @ polys[2] - G, rtol=1e-5, atol=1e-5)
+
+
+@pytest.mark.parametrize("kernel_type,K,expected", [
+    ("chebyshev", 2, 3), ("chebyshev", 3, 4), ("localpool", 2, 1),
+    ("random_walk_diffusion", 2, 3),  # preprocessor emits K+1 (quirk 3)
+])
+def test_dense_support_counts(kernel_type, K, expected):
+    gen = SupportGenerator(kernel_type, K)
+    stack = gen.process(_adj())
+    assert stack.shape[0] == expected
+
+
+def test_chebyshev_lambda_fixed2_matches_reference_behavior():
+    """torch.eig is dead on modern torch so the reference ALWAYS uses
+    lambda_max=2 (GCN.py:117-121): L~ = L - I."""
+    A = _adj()
+    gen = SupportGenerator("chebyshev", 2, lambda_max_mode="fixed2")
+    L = torch.eye(A.shape[0]) - symmetric_normalize(A)
+    torch.testing.assert_close(gen.generator(A), L - torch.eye(A.shape[0]),
+                               rtol=1e-6, atol=1e-6)
+
+
+def test_power_iteration_lmax():
+    A = _adj()
+    L = torch.eye(A.shape[0]) - symmetric_normalize(A)
+    lam = power_iteration_lmax(L, iters=500)
+    true_lam = torch.linalg.eigvalsh(L).max().item()
+    assert abs(lam - true_lam) < 1e-3
+
+
+def test_csr_roundtrip_and_dense_supports_match():
+    A = _adj()
+    gen = SupportGenerator("chebyshev", 2)
+    dense = gen.process(A)
+    csr = gen.process_csr(A)
+    assert csr.kind == "cheby" and csr.K_supports == 3
+    torch.testing.assert_close(csr.dense_generator(), gen.generator(A),
+                               rtol=1e-6, atol=1e-6)
+    torch.testing.assert_close(csr.dense_supports(), dense, rtol=1e-5, atol=1e-5)
+
+
+def test_csr_localpool():
+    A = _adj()
+    gen = SupportGenerator("localpool", 5)
+    csr = gen.process_csr(A)
+    assert csr.kind == "single" and csr.K_supports == 1
+    torch.testing.assert_close(csr.dense_supports(), gen.process(A),
+                               rtol=1e-6, atol=1e-6)
+
+
+def test_csr_cols_sorted_per_row():
+    csr = SupportGenerator("chebyshev", 2).process_csr(_adj())
+    for i in range(csr.n_nodes):
+        s, e = int(csr.row_ptr[i]), int(csr.row_ptr[i + 1])
+        cols = csr.col_idx[s:e]
+        assert (cols[1:] > cols[:-1]).all()

@@ -1,0 +1,120 @@
+"""End-to-end micro-training on synthetic data:
+checkpoint layout, early stopping, test metrics, Main.py wiring."""
+import os
+import subprocess
+import sys
+
+import numpy as np
+import pytest
+import torch
+from torch import nn, optim
+
+from stmgcn_amd.data import DataInput, DataGenerator, make_synthetic_dataset
+from stmgcn_amd.graph import SupportGenerator
+from stmgcn_amd.models import ST_MGCN
+from stmgcn_amd.train import ModelTrainer
+
+REPO = os.path.dirname(os.path.dirname(os.path.abspath(__file__)))
+
+
+def _setup(n_nodes=16, m=2, n_steps=24 * 30):
+    raw = make_synthetic_dataset(n_nodes=n_nodes, n_steps=n_steps, m_graphs=m)
+    di = DataInput(M_adj=m, data_dir="", norm_opt=True)
+    data = di.load_dict(raw)
+    gen = SupportGenerator("chebyshev", 2)
+    adjs = [gen.process(torch.from_numpy(data[k]).float())
+            for k in data if k.endswith("_adj")]
+    dgen = DataGenerator(dt=1, obs_len=(3, 1, 1),
+                         train_test_dates=["0101", "0630", "0701", "0731"])
+    loaders = dgen.get_data_loader(data, batch_size=32, device="cpu")
+    torch.manual_seed(0)
+    model = ST_MGCN(M=m, seq_len=5, n_nodes=n_nodes, input_dim=1,
+                    lstm_hidden_dim=16, lstm_num_layers=2, gcn_hidden_dim=16,
+                    sta_kernel_config={"kernel_type": "chebyshev", "K": 2})
+    return di, adjs, loaders, model
+
+
+def test_micro_train_checkpoint_and_test(tmp_path):
+    di, adjs, loaders, model = _setup()
+    trainer = ModelTrainer(model=model, loss=nn.MSELoss(), optimizer=optim.Adam,
+                           lr=2e-3, wd=1e-4, n_epochs=2,
+                           metrics_path=str(tmp_path / "metrics.jsonl"))
+    trainer.train(loaders, adjs, modes=["train", "validate"],
+                  model_dir=str(tmp_path))
+    ckpt_path = tmp_path / "ST_MGCN_best_model.pkl"
+    assert ckpt_path.exists()
+    ck = torch.load(ckpt_path, weights_only=False)
+    assert set(ck.keys()) == {"epoch", "state_dict"}
+    assert ck["epoch"] >= 1
+    assert len(ck["state_dict"]) == len(model.state_dict())
+
+    results = trainer.test(loaders, adjs, modes=["train", "test"],
+                           model_dir=str(tmp_path), data_class=di)
+    for mode in ["train", "test"]:
+        assert np.isfinite(results[mode]["RMSE"])
+        assert results[mode]["RMSE"] >= 0
+    assert (tmp_path / "metrics.jsonl").exists()
+
+
+def test_training_reduces_loss(tmp_path):
+    """Val loss after a few epochs must drop below the epoch-1 value
+    (sanity: the whole stack actually learns on periodic synthetic data)."""
+    di, adjs, loaders, model = _setup()
+    trainer = ModelTrainer(model=model, loss=nn.MSELoss(), optimizer=optim.Adam,
+                           lr=5e-3, wd=0.0, n_epochs=4)
+    losses = []
+
+    import stmgcn_amd.train.trainer as trmod
+    orig = trainer._allreduce_scalar
+    trainer._allreduce_scalar = lambda v: (losses.append(v), orig(v))[1]
+    trainer.train(loaders, adjs, modes=["train", "validate"], model_dir=str(tmp_path))
+    assert len(losses) == 4
+    assert min(losses[1:]) < losses[0]
+
+
+def test_early_stopping(tmp_path):
+    di, adjs, loaders, model = _setup(n_nodes=9, m=1, n_steps=24 * 10)
+    small = model.__class__(
+        M=1, seq_len=5, n_nodes=9, input_dim=1, lstm_hidden_dim=8,
+        lstm_num_layers=1, gcn_hidden_dim=8,
+        sta_kernel_config={"kernel_type": "chebyshev", "K": 2})
+    trainer = ModelTrainer(model=small, loss=nn.MSELoss(), optimizer=optim.Adam,
+                           lr=0.0, wd=0.0, n_epochs=50)
+    # force a strictly worsening val loss -> patience exhausts after epoch 1
+    seq = iter(range(1, 100))
+    trainer._allreduce_scalar = lambda v: float(next(seq))
+    trainer.train(loaders, adjs[:1], modes=["train", "validate"],
+                  model_dir=str(tmp_path), early_stopper=3)
+    ck = torch.load(tmp_path / "ST_MGCN_best_model.pkl", weights_only=False)
+    assert ck["epoch"] == 1  # only epoch 1 improved; stop fired before 50
+
+
+def test_trainer_rejects_unknown_model():
+    with pytest.raises(ValueError):
+        ModelTrainer(model=nn.Linear(3, 3), loss=nn.MSELoss(),
+                     optimizer=optim.Adam, lr=1e-3, wd=0, n_epochs=1)
+
+
+def test_main_cli_end_to_end(tmp_path):
+    """Main.py runs a tiny synthetic training + test on CPU."""
+    out = subprocess.run(
+        [sys.executable, os.path.join(REPO, "Main.py"),
+         "-device", "cpu", "--synthetic", "--preset", "cpu-small",
+         "--nodes", "9", "--epochs", "1", "--batch-size", "16",
+         "--model-dir", str(tmp_path)],
+        capture_output=True, text=True, timeout=600, cwd=REPO)
+    assert out.returncode == 0, out.stderr[-2000:]
+    assert "Training starts at:" in out.stdout
+    assert "true RMSE:" in out.stdout
+    assert (tmp_path / "ST_MGCN_best_model.pkl").exists()
+
+
+def test_main_cli_sparse_path(tmp_path):
+    out = subprocess.run(
+        [sys.executable, os.path.join(REPO, "Main.py"),
+         "-device", "cpu", "--synthetic", "--preset", "cpu-small",
+         "--nodes", "9", "--epochs", "1", "--batch-size", "16", "--sparse",
+         "--model-dir", str(tmp_path)],
+        capture_output=True, text=True, timeout=600, cwd=REPO)
+    assert out.returncode == 0, out.stderr[-2000:]
+    assert (tmp_path / "ST_MGCN_best_model.pkl").exists()
