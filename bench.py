@@ -1,0 +1,166 @@
+"""Flagship training benchmark — the driver contract.
+
+Measures the BASELINE.json north-star metric: whole-node train samples/sec of
+the 3-graph ST-MGCN at 1024 regions, seq_len 8, bf16, on synthetic
+region-demand tensors with random-init weights (the reference publishes no
+numbers; BASELINE.md's floor is the stock-PyTorch impl measured with
+--impl torch on the same config).
+
+Usage (driver): python -m torch.distributed.run --nnodes=1 --nproc-per-node N
+  --master-addr 127.0.0.1 --master-port P bench.py --gpus N --steps K --warmup W
+Single process: python bench.py [--gpus 1] [--steps K] [--warmup W]
+
+One training step = forward + loss + backward (+ DP all-reduce) + Adam step.
+Timed region: barrier + synchronize, K steps, barrier + synchronize; elapsed
+is MAX over ranks; rank 0 prints ONE JSON line.
+"""
+from __future__ import annotations
+
+import argparse
+import json
+import os
+import time
+
+import numpy as np
+import torch
+from torch import nn, optim
+
+from stmgcn_amd import PRESETS
+from stmgcn_amd.data.synthetic import make_synthetic_dataset
+from stmgcn_amd.graph import SupportGenerator
+from stmgcn_amd.models import build_model
+from stmgcn_amd.parallel import GradReducer, init_distributed, cleanup_distributed
+
+DTYPES = {"fp32": torch.float32, "bf16": torch.bfloat16, "fp16": torch.float16}
+
+
+def main():
+    p = argparse.ArgumentParser()
+    p.add_argument("--gpus", type=int, default=1)
+    p.add_argument("--steps", type=int, default=20)
+    p.add_argument("--warmup", type=int, default=5)
+    p.add_argument("--preset", type=str, default="bench-1024")
+    p.add_argument("--batch-size", type=int, default=None, help="per-rank batch")
+    p.add_argument("--impl", type=str, default=None, choices=["hip", "torch"],
+                   help="override STMGCN_IMPL (torch = stock-PyTorch floor)")
+    p.add_argument("--dtype", type=str, default=None, choices=sorted(DTYPES))
+    p.add_argument("--profile-trace", type=str, default=None,
+                   help="write a torch profiler trace to this path")
+    args = p.parse_args()
+
+    if args.impl:
+        os.environ["STMGCN_IMPL"] = args.impl
+    cfg = PRESETS[args.preset]
+    if args.batch_size:
+        cfg = cfg.replace(batch_size=args.batch_size)
+    if args.dtype:
+        cfg = cfg.replace(dtype=args.dtype)
+
+    env = init_distributed()
+    rank, world = env["rank"], env["world_size"]
+    use_gpu = torch.cuda.is_available()
+    device = torch.device(f"cuda:{env['local_rank']}") if use_gpu else torch.device("cpu")
+    if use_gpu:
+        torch.cuda.set_device(device)
+    dtype = DTYPES[cfg.dtype] if use_gpu else torch.float32
+
+    torch.manual_seed(1234 + rank)
+    # ---- synthetic inputs of the target shape (weak scaling: per-GPU fixed) --
+    B, T, N, C = cfg.batch_size, cfg.seq_len, cfg.n_nodes, cfg.input_dim
+    raw = make_synthetic_dataset(n_nodes=N, n_steps=max(T * 8, 64), m_graphs=cfg.m_graphs,
+                                 seed=7, day_timesteps=1)
+    gen = SupportGenerator(cfg.kernel_type, cfg.cheby_K, cfg.lambda_max_mode)
+    adj_keys = [k for k in raw if k.endswith("_adj")]
+    if use_gpu and os.environ.get("STMGCN_IMPL", "hip") == "hip":
+        adjs = [gen.process_csr(torch.from_numpy(raw[k]).float()).to(device) for k in adj_keys]
+    else:
+        adjs = [gen.process(torch.from_numpy(raw[k]).float()).to(device=device, dtype=dtype)
+                for k in adj_keys]
+    x = torch.randn(B, T, N, C, device=device, dtype=dtype)
+    y = torch.randn(B, N, C, device=device, dtype=dtype)
+
+    # ---- model + optimizer --------------------------------------------------
+    model = build_model(cfg).to(device=device, dtype=dtype)
+    criterion = {"MSE": nn.MSELoss(), "MAE": nn.L1Loss(), "Huber": nn.SmoothL1Loss()}[cfg.loss]
+    reducer = GradReducer(model) if world > 1 else None
+    opt = optim.Adam(model.parameters(), lr=cfg.lr, weight_decay=cfg.weight_decay)
+
+    def step():
+        if reducer is not None:
+            reducer.zero_grad()
+        else:
+            opt.zero_grad(set_to_none=True)
+        loss = criterion(model(x, adjs), y)
+        loss.backward()
+        if reducer is not None:
+            reducer.reduce()
+        opt.step()
+        return loss
+
+    def barrier_sync():
+        if world > 1:
+            torch.distributed.barrier()
+        if use_gpu:
+            torch.cuda.synchronize()
+
+    for _ in range(args.warmup):
+        step()
+    barrier_sync()
+
+    prof_ctx = None
+    if args.profile_trace:
+        prof_ctx = torch.profiler.profile(
+            activities=[torch.profiler.ProfilerActivity.CPU,
+                        torch.profiler.ProfilerActivity.CUDA])
+        prof_ctx.__enter__()
+
+    t0 = time.perf_counter()
+    for _ in range(args.steps):
+        step()
+    barrier_sync()
+    elapsed = time.perf_counter() - t0
+
+    if prof_ctx is not None:
+        prof_ctx.__exit__(None, None, None)
+        if rank == 0:
+            prof_ctx.export_chrome_trace(args.profile_trace)
+
+    # max over ranks
+    if world > 1:
+        t = torch.tensor([elapsed], dtype=torch.float64,
+                         device=device if torch.distributed.get_backend() == "nccl" else "cpu")
+        torch.distributed.all_reduce(t, op=torch.distributed.ReduceOp.MAX)
+        elapsed = float(t.item())
+
+    n_gpus = world if use_gpu else world  # ranks == GPUs in the driver launch
+    samples_per_sec = n_gpus * B * args.steps / elapsed
+    if rank == 0:
+        rec = {
+            "metric": "train_samples_per_sec",
+            "value": samples_per_sec,
+            "unit": "samples/s",
+            "n_gpus": n_gpus,
+            "steps": args.steps,
+            "warmup": args.warmup,
+            "ms_per_step": elapsed / args.steps * 1e3,
+            "higher_is_better": True,
+            "scaling": "weak",
+            "vs_baseline": None,
+            "dtype": cfg.dtype if use_gpu else "fp32",
+            "data": "synthetic",
+            "config": {
+                "model": model.__class__.__name__,
+                "preset": args.preset,
+                "n_nodes": N, "seq_len": T, "m_graphs": cfg.m_graphs,
+                "lstm_hidden": cfg.lstm_hidden_dim, "gcn_hidden": cfg.gcn_hidden_dim,
+                "global_batch": B * n_gpus,
+                "parallelism": f"dp{n_gpus}",
+                "impl": os.environ.get("STMGCN_IMPL", "hip") if use_gpu else "torch-cpu",
+            },
+        }
+        print(json.dumps(rec))
+    cleanup_distributed()
+
+
+if __name__ == "__main__":
+    main()

@@ -1,0 +1,34 @@
+// Common device helpers for the stmgcn_amd CDNA4 (gfx950) kernels.
+#pragma once
+#include <hip/hip_runtime.h>
+#include <hip/hip_bf16.h>
+#include <hip/hip_fp16.h>
+
+#define STM_WAVE 64  // CDNA wavefront width (never 32)
+
+// dtype codes shared with the python side
+enum StmDtype : int { STM_F32 = 0, STM_BF16 = 1, STM_F16 = 2 };
+
+template <typename T> __device__ __forceinline__ float toF(T v);
+template <> __device__ __forceinline__ float toF<float>(float v) { return v; }
+template <> __device__ __forceinline__ float toF<__hip_bfloat16>(__hip_bfloat16 v) {
+  return __bfloat162float(v);
+}
+template <> __device__ __forceinline__ float toF<__half>(__half v) { return __half2float(v); }
+
+template <typename T> __device__ __forceinline__ T fromF(float v);
+template <> __device__ __forceinline__ float fromF<float>(float v) { return v; }
+template <> __device__ __forceinline__ __hip_bfloat16 fromF<__hip_bfloat16>(float v) {
+  return __float2bfloat16(v);
+}
+template <> __device__ __forceinline__ __half fromF<__half>(float v) { return __float2half(v); }
+
+__device__ __forceinline__ float stm_sigmoid(float x) { return 1.0f / (1.0f + __expf(-x)); }
+
+#define STM_CHECK_HIP(expr)                                                     \
+  do {                                                                          \
+    hipError_t _e = (expr);                                                     \
+    if (_e != hipSuccess) {                                                     \
+      printf("HIP error %s at %s:%d\n", hipGetErrorString(_e), __FILE__, __LINE__); \
+    }                                                                           \
+  } while (0)

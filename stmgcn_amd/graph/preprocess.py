@@ -72,6 +72,10 @@ class CSRSupport:
     kind == "cheby":  supports are T_0=I, T_1=G, T_k = 2 G T_{k-1} - T_{k-2},
                       K_s = K+1 of them (chebyshev & rw-diffusion families).
     kind == "single": one support, the matrix G itself (localpool).
+
+    Carries BOTH G and G^T in CSR: backward applies T_k(G)^T = T_k(G^T)
+    (rw-diffusion's P^T is asymmetric; for the symmetric chebyshev/localpool
+    generators the two are identical but stored uniformly).
     """
     row_ptr: torch.Tensor   # int32 (N+1,)
     col_idx: torch.Tensor   # int32 (nnz,)
@@ -79,10 +83,17 @@ class CSRSupport:
     n_nodes: int
     K_supports: int
     kind: str               # "cheby" | "single"
+    row_ptr_t: Optional[torch.Tensor] = None   # CSR of G^T
+    col_idx_t: Optional[torch.Tensor] = None
+    vals_t: Optional[torch.Tensor] = None
 
     def to(self, device) -> "CSRSupport":
         return CSRSupport(self.row_ptr.to(device), self.col_idx.to(device),
-                          self.vals.to(device), self.n_nodes, self.K_supports, self.kind)
+                          self.vals.to(device), self.n_nodes, self.K_supports,
+                          self.kind,
+                          None if self.row_ptr_t is None else self.row_ptr_t.to(device),
+                          None if self.col_idx_t is None else self.col_idx_t.to(device),
+                          None if self.vals_t is None else self.vals_t.to(device))
 
     @property
     def nnz(self) -> int:
@@ -114,9 +125,7 @@ def chebyshev_polynomials(G: torch.Tensor, K: int) -> List[torch.Tensor]:
     return out
 
 
-def dense_to_csr(G: torch.Tensor, prune_eps: float = 0.0) -> CSRSupport:
-    """Dense (N,N) -> CSR (int32 indices, fp32 values). Rows kept sorted by
-    column for coalesced in-kernel gathers."""
+def _csr_arrays(G: torch.Tensor, prune_eps: float):
     mask = G.abs() > prune_eps
     N = G.shape[0]
     counts = mask.sum(dim=1, dtype=torch.int32)
@@ -125,7 +134,15 @@ def dense_to_csr(G: torch.Tensor, prune_eps: float = 0.0) -> CSRSupport:
     idx = mask.nonzero(as_tuple=False)  # sorted row-major -> cols sorted per row
     col_idx = idx[:, 1].to(torch.int32).contiguous()
     vals = G[mask].to(torch.float32).contiguous()
-    return CSRSupport(row_ptr, col_idx, vals, N, 0, "")
+    return row_ptr, col_idx, vals
+
+
+def dense_to_csr(G: torch.Tensor, prune_eps: float = 0.0) -> CSRSupport:
+    """Dense (N,N) -> CSR of G and of G^T (int32 indices, fp32 values).
+    Rows kept sorted by column for coalesced in-kernel gathers."""
+    rp, ci, v = _csr_arrays(G, prune_eps)
+    rpt, cit, vt = _csr_arrays(G.T.contiguous(), prune_eps)
+    return CSRSupport(rp, ci, v, G.shape[0], 0, "", rpt, cit, vt)
 
 
 class SupportGenerator:

@@ -18,6 +18,15 @@ _HIP = None
 _HIP_ERR: Optional[str] = None
 
 
+def impl_mode() -> str:
+    """"hip" (default): GPU tensors require the HIP extension, fail-loud.
+    "torch": explicit opt-in eager-PyTorch execution on GPU — used ONLY to
+    measure the stock-PyTorch comparison floor and for A/B
+    numerics debugging. Set STMGCN_IMPL=torch to opt in."""
+    import os
+    return os.environ.get("STMGCN_IMPL", "hip")
+
+
 def _load_hip():
     global _HIP, _HIP_ERR
     if _HIP is not None or _HIP_ERR is not None:
@@ -50,7 +59,7 @@ def gconv_mix(A, x: torch.Tensor, W: torch.Tensor, b: Optional[torch.Tensor],
     A is either a dense (K, N, N) support stack (reference parity path,
     GCN.py:24-43) or a CSRSupport (in-kernel Chebyshev recurrence path)."""
     if isinstance(A, CSRSupport):
-        if x.is_cuda:
+        if x.is_cuda and impl_mode() == "hip":
             from .hip_ops import ChebGconvFn
             return ChebGconvFn.apply(x, W, b, A, activation)
         return ref.gconv_mix_csr(A, x, W, b, activation)
@@ -60,7 +69,7 @@ def gconv_mix(A, x: torch.Tensor, W: torch.Tensor, b: Optional[torch.Tensor],
 def contextual_gate(obs_seq: torch.Tensor, gconv_out: torch.Tensor,
                     fc_weight: torch.Tensor, fc_bias: torch.Tensor) -> torch.Tensor:
     """Contextual gating (eqs.6-9, weight-tied FC — reference STMGCN.py:36-44)."""
-    if obs_seq.is_cuda:
+    if obs_seq.is_cuda and impl_mode() == "hip":
         from .hip_ops import contextual_gate_hip
         return contextual_gate_hip(obs_seq, gconv_out, fc_weight, fc_bias)
     return ref.contextual_gate(obs_seq, gconv_out, fc_weight, fc_bias)
@@ -71,7 +80,7 @@ def rnn_forward(cell: str, x: torch.Tensor, weights: List[torch.Tensor],
                 return_sequences: bool = False) -> torch.Tensor:
     """Multi-layer LSTM/GRU over (B*N, T, C) — the dominant-FLOP op
     (reference STMGCN.py:47-50). GPU: persistent fused HIP kernel."""
-    if x.is_cuda:
+    if x.is_cuda and impl_mode() == "hip":
         from .hip_ops import FusedRNNFn
         return FusedRNNFn.apply(cell, x, h0, c0, return_sequences, *weights)
     if cell == "lstm":
@@ -84,7 +93,7 @@ def rnn_forward(cell: str, x: torch.Tensor, weights: List[torch.Tensor],
 def branch_fuse_head(branch_feats: List[torch.Tensor], fc_weight: torch.Tensor,
                      fc_bias: torch.Tensor) -> torch.Tensor:
     """Sum over M branches + FC head (reference STMGCN.py:116-118)."""
-    if branch_feats[0].is_cuda:
+    if branch_feats[0].is_cuda and impl_mode() == "hip":
         from .hip_ops import branch_fuse_head_hip
         return branch_fuse_head_hip(branch_feats, fc_weight, fc_bias)
     return ref.branch_fuse_head(branch_feats, fc_weight, fc_bias)
