@@ -131,7 +131,104 @@ at::Tensor cheb_combine(at::Tensor U, at::Tensor rowptr, at::Tensor colidx,
   return Z;
 }
 
+// ---------------------------------------------------------------------------
+// Fused LSTM: forward kernel + dgrad kernel; weight grads
+// are computed on the python side as plain library GEMMs over the dA stream.
+
+extern "C" void stmgcn_lstm_fwd(void* stream, int dtype, const void* x,
+                                void* out, void* hseq_g, void* cseq_g,
+                                void* gates_g, const void** w_ih,
+                                const void** w_hh, const void** b_ih,
+                                const void** b_hh, int S, int Tst, int L,
+                                int cin, int ret_seq);
+extern "C" void stmgcn_lstm_bwd(void* stream, int dtype, const void* dout,
+                                const void* x, const void* cseq_g,
+                                const void* gates_g, const void** w_ihT,
+                                const void** w_hhT, void* dx, void* dA_g,
+                                int S, int Tst, int L, int cin, int ret_seq);
+extern "C" void stmgcn_mfma_probe(void* stream, const void* A, const void* B,
+                                  void* D);
+
+static constexpr int kSeqTile = 64;
+static constexpr int kH = 64;
+
+std::vector<at::Tensor> lstm_fwd(at::Tensor x, std::vector<at::Tensor> w_ih,
+                                 std::vector<at::Tensor> w_hh,
+                                 std::vector<at::Tensor> b_ih,
+                                 std::vector<at::Tensor> b_hh, bool ret_seq,
+                                 bool training) {
+  TORCH_CHECK(x.is_cuda() && x.dim() == 3 && x.is_contiguous());
+  TORCH_CHECK(x.scalar_type() == at::kBFloat16 || x.scalar_type() == at::kHalf,
+              "fused LSTM serves bf16/f16 (fp32 runs the torch path)");
+  const int S = x.size(0), Tst = x.size(1), cin = x.size(2);
+  const int L = (int)w_ih.size();
+  TORCH_CHECK(L >= 1 && L <= 8 && Tst <= 16);
+  TORCH_CHECK(cin == 1 || cin == kH, "C_in must be 1 or 64");
+  for (int l = 0; l < L; ++l) {
+    TORCH_CHECK(w_hh[l].size(0) == 4 * kH && w_hh[l].size(1) == kH,
+                "hidden dim must be 64");
+    TORCH_CHECK(w_ih[l].is_contiguous() && w_hh[l].is_contiguous());
+    TORCH_CHECK(b_ih[l].scalar_type() == at::kFloat &&
+                b_hh[l].scalar_type() == at::kFloat);
+  }
+  const long nblk = (S + kSeqTile - 1) / kSeqTile;
+  const long S_pad = nblk * kSeqTile;
+  auto out = ret_seq ? at::empty({S, Tst, kH}, x.options())
+                     : at::empty({S, kH}, x.options());
+  at::Tensor hseq, cseq, gates;
+  const void *wi[8], *wh[8], *bi[8], *bh[8];
+  for (int l = 0; l < L; ++l) {
+    wi[l] = w_ih[l].data_ptr(); wh[l] = w_hh[l].data_ptr();
+    bi[l] = b_ih[l].data_ptr(); bh[l] = b_hh[l].data_ptr();
+  }
+  void *hp = nullptr, *cp = nullptr, *gp = nullptr;
+  if (training) {
+    hseq = at::empty({L, Tst, S_pad, kH}, x.options());
+    cseq = at::empty({L, Tst, S_pad * kH}, x.options().dtype(at::kFloat));
+    gates = at::empty({L, Tst, S_pad * 4 * kH}, x.options());
+    hp = hseq.data_ptr(); cp = cseq.data_ptr(); gp = gates.data_ptr();
+  }
+  stmgcn_lstm_fwd(stream(), dtype_code(x), x.data_ptr(), out.data_ptr(), hp,
+                  cp, gp, wi, wh, bi, bh, S, Tst, L, cin, ret_seq ? 1 : 0);
+  if (!training) return {out};
+  return {out, hseq, cseq, gates};
+}
+
+std::vector<at::Tensor> lstm_bwd(at::Tensor dout, at::Tensor x,
+                                 at::Tensor cseq, at::Tensor gates,
+                                 std::vector<at::Tensor> w_ihT,
+                                 std::vector<at::Tensor> w_hhT, bool ret_seq) {
+  TORCH_CHECK(dout.is_cuda() && x.is_contiguous());
+  dout = dout.contiguous();
+  const int S = x.size(0), Tst = x.size(1), cin = x.size(2);
+  const int L = (int)w_ihT.size();
+  const long nblk = (S + kSeqTile - 1) / kSeqTile;
+  const long S_pad = nblk * kSeqTile;
+  auto dx = at::empty_like(x);
+  auto dA = at::empty({L, Tst, S_pad, 4 * kH}, x.options());
+  const void *wi[8], *wh[8];
+  for (int l = 0; l < L; ++l) {
+    TORCH_CHECK(w_ihT[l].is_contiguous() && w_hhT[l].is_contiguous());
+    wi[l] = w_ihT[l].data_ptr(); wh[l] = w_hhT[l].data_ptr();
+  }
+  stmgcn_lstm_bwd(stream(), dtype_code(x), dout.data_ptr(), x.data_ptr(),
+                  cseq.data_ptr(), gates.data_ptr(), wi, wh, dx.data_ptr(),
+                  dA.data_ptr(), S, Tst, L, cin, ret_seq ? 1 : 0);
+  return {dx, dA};
+}
+
+at::Tensor mfma_probe(at::Tensor A, at::Tensor B) {
+  TORCH_CHECK(A.is_cuda() && A.scalar_type() == at::kBFloat16);
+  auto D = at::empty({16, 16}, A.options().dtype(at::kFloat));
+  stmgcn_mfma_probe(stream(), A.contiguous().data_ptr(),
+                    B.contiguous().data_ptr(), D.data_ptr());
+  return D;
+}
+
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
+  m.def("lstm_fwd", &lstm_fwd, "Fused multi-layer LSTM forward (persistent)");
+  m.def("lstm_bwd", &lstm_bwd, "Fused LSTM dgrad (BPTT in-kernel)");
+  m.def("mfma_probe", &mfma_probe, "16x16x32 bf16 MFMA fragment-layout probe");
   m.def("cheb_apply", &cheb_apply,
         "Support stack S[b,n,k,c] = (T_k(G) x)[b,n,c] via in-kernel recurrence");
   m.def("cheb_combine", &cheb_combine,

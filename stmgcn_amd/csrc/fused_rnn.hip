@@ -1,0 +1,588 @@
+// Persistent fused multi-layer LSTM/GRU for CDNA4 (gfx950) — SURVEY K5/K6.
+//
+// Replaces the reference's nn.LSTM over the flattened (B*N, T, C) sequence
+// batch (STMGCN.py:21-22,47-50). One kernel runs ALL layers and ALL
+// timesteps: each 256-thread workgroup owns a 64-sequence tile, stages the
+// evolving per-timestep hidden sequence in LDS (XOR-swizzled for
+// conflict-free ds_read_b128 MFMA fragment reads), computes the recurrent +
+// input projections with v_mfma_f32_16x16x32_bf16 matrix cores (fp32
+// accumulate), and applies the gate nonlinearities in the MFMA accumulator
+// fragment layout, so hidden state never leaves the CU between timesteps.
+//
+// Geometry per workgroup (H = 64 fixed):
+//   4 waves; wave w owns gate-channel slice [16w, 16w+16) of each of the
+//   4 (LSTM) / 3 (GRU) gate types -> i/f/g/o for one (seq,channel) pair land
+//   in the SAME lane's accumulators, making the cell update lane-local.
+//   M = 64 sequences = 4 MFMA row-tiles; K = H (+C for layers > 0) chunked
+//   by 32. Weights are loaded once per layer into VGPR B-fragments
+//   (contiguous 16B per lane from the (4H, H|C) row-major weight).
+//
+// Saves for BPTT (training): per (layer, t): post-activation gates + cell
+// state in fragment-native order (coalesced 16B/lane), and the hidden
+// sequence in natural (S_pad, H) order staged through LDS.
+//
+// fp32/fp64 paths are not served here: bf16/f16 in, fp32 accumulate.
+
+#include "common.h"
+
+typedef __attribute__((ext_vector_type(8))) __bf16 bf16x8;
+typedef __attribute__((ext_vector_type(8))) _Float16 f16x8;
+typedef __attribute__((ext_vector_type(4))) float f32x4;
+
+#define SEQ_TILE 64
+#define RNN_H 64
+#define MAX_LAYERS 8
+#define MAX_T 16
+
+struct RnnPtrs {
+  const void* w_ih[MAX_LAYERS];   // (G*H, C_l) row-major, model dtype
+  const void* w_hh[MAX_LAYERS];   // (G*H, H)
+  const void* b_ih[MAX_LAYERS];   // (G*H,) fp32
+  const void* b_hh[MAX_LAYERS];   // (G*H,) fp32
+};
+
+// ---------------------------------------------------------------------------
+// LDS swizzle: element (row s in [0,64), channel c in [0,64)) of one 8 KiB
+// timestep slot. Byte = s*128 + c*2, XOR'd so the 16 lanes of a ds_read_b128
+// group (consecutive rows, same channel block) hit distinct banks.
+__device__ __forceinline__ int lds_swz(int s, int cbyte) {
+  return s * 128 + (cbyte ^ ((s & 7) << 4));
+}
+
+template <typename T> struct Frag8;
+template <> struct Frag8<__hip_bfloat16> { using type = bf16x8; using elem = __bf16; };
+template <> struct Frag8<__half> { using type = f16x8; using elem = _Float16; };
+
+__device__ __forceinline__ float elemF(__bf16 v) { return (float)v; }
+__device__ __forceinline__ float elemF(_Float16 v) { return (float)v; }
+
+__device__ __forceinline__ f32x4 mfma16x16x32(bf16x8 a, bf16x8 b, f32x4 c) {
+  return __builtin_amdgcn_mfma_f32_16x16x32_bf16(a, b, c, 0, 0, 0);
+}
+__device__ __forceinline__ f32x4 mfma16x16x32(f16x8 a, f16x8 b, f32x4 c) {
+  return __builtin_amdgcn_mfma_f32_16x16x32_f16(a, b, c, 0, 0, 0);
+}
+
+// ---------------------------------------------------------------------------
+// MFMA layout probe (validated on hardware by tests/test_gpu_kernels.py):
+// D[16,16] = A[16,32] @ B[32,16] with the assumed fragment maps:
+//   A: lane l holds A[l%16][(l/16)*8 + j], j = 0..7
+//   B: lane l holds B[(l/16)*8 + j][l%16]
+//   C/D: lane l holds D[(l/16)*4 + r][l%16], r = 0..3
+extern "C" __global__ void mfma_probe_kernel(const __hip_bfloat16* A,
+                                             const __hip_bfloat16* B, float* D) {
+  const int l = threadIdx.x;
+  bf16x8 a, b;
+  #pragma unroll
+      for (int j = 0; j < 8; ++j) {
+    a[j] = *(const __bf16*)&A[(l % 16) * 32 + (l / 16) * 8 + j];
+    b[j] = *(const __bf16*)&B[((l / 16) * 8 + j) * 16 + (l % 16)];
+  }
+  f32x4 c = {0.f, 0.f, 0.f, 0.f};
+  c = mfma16x16x32(a, b, c);
+  #pragma unroll
+      for (int r = 0; r < 4; ++r) D[((l / 16) * 4 + r) * 16 + (l % 16)] = c[r];
+}
+
+// ---------------------------------------------------------------------------
+// Fused LSTM forward.
+//
+// Template: T = element type (bf16/f16), CIN1 = first-layer input_dim == 1
+// (base ST-MGCN; otherwise C_in == H, the stacked deep-variant blocks).
+//
+// LDS map (dynamic): hseq slots [T_steps][64][64] T-typed, swizzled;
+// then xbuf: CIN1 ? [T_steps][64] : [T_steps][64][64] (swizzled).
+template <typename T, bool CIN1>
+__global__ void __launch_bounds__(256, 1)
+lstm_fwd_kernel(const T* __restrict__ x,    // (S, Tst, C_in)
+                T* __restrict__ out,        // (S, H) or (S, Tst, H)
+                T* __restrict__ hseq_g,     // (L, Tst, S_pad, H) or null
+                float* __restrict__ cseq_g, // (L, Tst, S_pad*H) frag-native
+                T* __restrict__ gates_g,    // (L, Tst, S_pad*4H) frag-native
+                RnnPtrs ptrs,
+                int S, int Tst, int L, int ret_seq) {
+  using frag = typename Frag8<T>::type;
+  extern __shared__ char lds[];
+  const int s0 = blockIdx.x * SEQ_TILE;
+  const int wv = threadIdx.x >> 6;
+  const int lane = threadIdx.x & 63;
+  const int l16 = lane & 15;
+  const int lgrp = lane >> 4;
+  const long S_pad = (long)gridDim.x * SEQ_TILE;
+
+  char* hseq = lds;                                   // Tst * 8192 bytes
+  char* xbuf = lds + Tst * 8192;
+
+  // ---- stage input x into LDS --------------------------------------------
+  if (CIN1) {
+    // x (S, Tst, 1) -> xbuf[t*64 + s] (T-typed)
+    for (int i = threadIdx.x; i < SEQ_TILE * Tst; i += 256) {
+      const int s = i & 63, t = i >> 6;
+      T v = fromF<T>(0.f);
+      if (s0 + s < S) v = x[(long)(s0 + s) * Tst + t];
+      ((T*)xbuf)[t * 64 + s] = v;
+    }
+  } else {
+    // x (S, Tst, 64) -> xbuf slot t swizzled, 16B pieces
+    for (int i = threadIdx.x; i < SEQ_TILE * Tst * 8; i += 256) {
+      const int c8 = i & 7, s = (i >> 3) & 63, t = i >> 9;
+      frag v = {};
+      if (s0 + s < S)
+        v = *(const frag*)&x[((long)(s0 + s) * Tst + t) * RNN_H + c8 * 8];
+      *(frag*)&xbuf[t * 8192 + lds_swz(s, c8 * 16)] = v;
+    }
+  }
+  __syncthreads();
+
+  // per-lane fragment row/col constants
+  //   acc row r(m, reg) = 16m + 4*lgrp + reg ; acc col = l16 (+16w for h ch)
+  const int hch = 16 * wv + l16;                      // this lane's h channel
+
+  for (int layer = 0; layer < L; ++layer) {
+    const int cin = (layer == 0) ? (CIN1 ? 1 : RNN_H) : RNN_H;
+    const T* Whh = (const T*)ptrs.w_hh[layer];
+    const T* Wih = (const T*)ptrs.w_ih[layer];
+    const float* bih = (const float*)ptrs.b_ih[layer];
+    const float* bhh = (const float*)ptrs.b_hh[layer];
+
+    // B-fragments for the recurrent GEMM: b_hh[q][kk]; W row g = q*64 + hch,
+    // cols kk*32 + lgrp*8 .. +8 -> contiguous 16B in the (4H, H) weight.
+    frag bhfrag[4][2];
+    #pragma unroll
+      for (int q = 0; q < 4; ++q)
+      #pragma unroll
+      for (int kk = 0; kk < 2; ++kk)
+        bhfrag[q][kk] = *(const frag*)&Whh[(q * 64 + hch) * RNN_H + kk * 32 + lgrp * 8];
+    // input-projection B-fragments (dense layers only)
+    frag bxfrag[4][2];
+    if (!CIN1 || layer > 0)
+      #pragma unroll
+      for (int q = 0; q < 4; ++q)
+        #pragma unroll
+      for (int kk = 0; kk < 2; ++kk)
+          bxfrag[q][kk] = *(const frag*)&Wih[(q * 64 + hch) * RNN_H + kk * 32 + lgrp * 8];
+    // layer-0 scalar input weights (CIN1): W_ih[g][0]
+    float wih0[4];
+    float bias[4];
+    #pragma unroll
+      for (int q = 0; q < 4; ++q) {
+      const int g = q * 64 + hch;
+      bias[q] = bih[g] + bhh[g];
+      if (CIN1 && layer == 0) wih0[q] = toF<T>(Wih[g]);
+    }
+
+    float c_state[4][4];  // [m][reg]
+    #pragma unroll
+      for (int m = 0; m < 4; ++m)
+      #pragma unroll
+      for (int r = 0; r < 4; ++r) c_state[m][r] = 0.f;
+
+    for (int t = 0; t < Tst; ++t) {
+      f32x4 acc[4][4];  // [m][q]
+      #pragma unroll
+      for (int m = 0; m < 4; ++m)
+        #pragma unroll
+      for (int q = 0; q < 4; ++q) acc[m][q] = f32x4{0.f, 0.f, 0.f, 0.f};
+
+      // recurrent term: h_{t-1} from hseq slot t-1 (zero at t == 0)
+      if (t > 0) {
+        char* slot = hseq + (t - 1) * 8192;
+        #pragma unroll
+      for (int m = 0; m < 4; ++m) {
+          const int row = 16 * m + l16;
+          #pragma unroll
+      for (int kk = 0; kk < 2; ++kk) {
+            frag a = *(const frag*)&slot[lds_swz(row, (kk * 32 + lgrp * 8) * 2)];
+            #pragma unroll
+      for (int q = 0; q < 4; ++q)
+              acc[m][q] = mfma16x16x32(a, bhfrag[q][kk], acc[m][q]);
+          }
+        }
+      }
+      // input term
+      if (!CIN1 || layer > 0) {
+        char* src = (layer == 0) ? (xbuf + t * 8192) : (hseq + t * 8192);
+        #pragma unroll
+      for (int m = 0; m < 4; ++m) {
+          const int row = 16 * m + l16;
+          #pragma unroll
+      for (int kk = 0; kk < 2; ++kk) {
+            frag a = *(const frag*)&src[lds_swz(row, (kk * 32 + lgrp * 8) * 2)];
+            #pragma unroll
+      for (int q = 0; q < 4; ++q)
+              acc[m][q] = mfma16x16x32(a, bxfrag[q][kk], acc[m][q]);
+          }
+        }
+      }
+
+      // pointwise cell update in fragment layout (+ scalar-input term)
+      T hval[4][4];     // [m][reg] this lane's h outputs (ch = hch)
+      T gsave[4][16];   // [m][i f g o][reg] post-activation gates
+      #pragma unroll
+      for (int m = 0; m < 4; ++m) {
+        float xv[4];
+        if (CIN1 && layer == 0)
+          #pragma unroll
+      for (int r = 0; r < 4; ++r)
+            xv[r] = toF<T>(((const T*)xbuf)[t * 64 + 16 * m + 4 * lgrp + r]);
+        #pragma unroll
+      for (int r = 0; r < 4; ++r) {
+          float gi = acc[m][0][r] + bias[0];
+          float gf = acc[m][1][r] + bias[1];
+          float gg = acc[m][2][r] + bias[2];
+          float go = acc[m][3][r] + bias[3];
+          if (CIN1 && layer == 0) {
+            gi += xv[r] * wih0[0]; gf += xv[r] * wih0[1];
+            gg += xv[r] * wih0[2]; go += xv[r] * wih0[3];
+          }
+          const float i_ = stm_sigmoid(gi), f_ = stm_sigmoid(gf);
+          const float g_ = tanhf(gg), o_ = stm_sigmoid(go);
+          const float c_ = f_ * c_state[m][r] + i_ * g_;
+          c_state[m][r] = c_;
+          const float h_ = o_ * tanhf(c_);
+          hval[m][r] = fromF<T>(h_);
+          gsave[m][0 * 4 + r] = fromF<T>(i_);
+          gsave[m][1 * 4 + r] = fromF<T>(f_);
+          gsave[m][2 * 4 + r] = fromF<T>(g_);
+          gsave[m][3 * 4 + r] = fromF<T>(o_);
+        }
+      }
+
+      // wait: all waves finished reading slot t (input term) before overwrite
+      __syncthreads();
+      {
+        char* slot = hseq + t * 8192;
+        #pragma unroll
+      for (int m = 0; m < 4; ++m)
+          #pragma unroll
+      for (int r = 0; r < 4; ++r) {
+            const int row = 16 * m + 4 * lgrp + r;
+            *(T*)&slot[lds_swz(row, hch * 2)] = hval[m][r];
+          }
+      }
+      __syncthreads();
+
+      // training saves: gates + cell, fragment-native (16B contiguous/lane)
+      if (gates_g) {
+        const long base = ((long)layer * Tst + t);
+        // gates: (L,Tst, S_pad*4H) as [wave][m][lane][16] T
+        T* gp = gates_g + base * (S_pad * 4 * RNN_H)
+                + (long)blockIdx.x * (SEQ_TILE * 4 * RNN_H)
+                + ((wv * 4) * 64) * 16;
+        #pragma unroll
+      for (int m = 0; m < 4; ++m)
+          *(((frag*)(gp + (m * 64 + lane) * 16)) + 0) = *(frag*)&gsave[m][0],
+          *(((frag*)(gp + (m * 64 + lane) * 16)) + 1) = *(frag*)&gsave[m][8];
+        // cell: (L,Tst, S_pad*H) fp32 as [wave][m][lane][4]
+        float* cp = cseq_g + base * (S_pad * RNN_H)
+                    + (long)blockIdx.x * (SEQ_TILE * RNN_H) + (wv * 4) * 64 * 4;
+        #pragma unroll
+      for (int m = 0; m < 4; ++m)
+          *(f32x4*)(cp + (m * 64 + lane) * 4) =
+              f32x4{c_state[m][0], c_state[m][1], c_state[m][2], c_state[m][3]};
+        // hseq natural layout copy of slot t (also next layer's input source)
+        T* hp = hseq_g + base * (S_pad * RNN_H) + (long)(s0)*RNN_H;
+        char* slot = hseq + t * 8192;
+        for (int i = threadIdx.x; i < SEQ_TILE * 8; i += 256) {
+          const int c8 = i & 7, s = i >> 3;
+          *(frag*)&hp[s * RNN_H + c8 * 8] = *(frag*)&slot[lds_swz(s, c8 * 16)];
+        }
+        __syncthreads();
+      }
+    }
+  }
+
+  // ---- output ------------------------------------------------------------
+  if (ret_seq) {
+    for (int i = threadIdx.x; i < SEQ_TILE * Tst * 8; i += 256) {
+      const int c8 = i & 7, s = (i >> 3) & 63, t = i >> 9;
+      if (s0 + s < S)
+        *(frag*)&out[((long)(s0 + s) * Tst + t) * RNN_H + c8 * 8] =
+            *(frag*)&hseq[t * 8192 + lds_swz(s, c8 * 16)];
+    }
+  } else {
+    char* slot = hseq + (Tst - 1) * 8192;
+    for (int i = threadIdx.x; i < SEQ_TILE * 8; i += 256) {
+      const int c8 = i & 7, s = i >> 3;
+      if (s0 + s < S)
+        *(frag*)&out[(long)(s0 + s) * RNN_H + c8 * 8] =
+            *(frag*)&slot[lds_swz(s, c8 * 16)];
+    }
+  }
+}
+
+template <typename T>
+void launch_fwd(hipStream_t stream, const void* x, void* out, void* hseq_g,
+                void* cseq_g, void* gates_g, const RnnPtrs& ptrs, int S,
+                int Tst, int L, int cin, int ret_seq) {
+  const int nblk = (S + SEQ_TILE - 1) / SEQ_TILE;
+  const bool cin1 = (cin == 1);
+  const size_t lds_bytes = (size_t)Tst * 8192 + (cin1 ? Tst * 64 * sizeof(T) : (size_t)Tst * 8192);
+  if (cin1)
+    hipLaunchKernelGGL((lstm_fwd_kernel<T, true>), dim3(nblk), dim3(256),
+                       lds_bytes, stream, (const T*)x, (T*)out, (T*)hseq_g,
+                       (float*)cseq_g, (T*)gates_g, ptrs, S, Tst, L, ret_seq);
+  else
+    hipLaunchKernelGGL((lstm_fwd_kernel<T, false>), dim3(nblk), dim3(256),
+                       lds_bytes, stream, (const T*)x, (T*)out, (T*)hseq_g,
+                       (float*)cseq_g, (T*)gates_g, ptrs, S, Tst, L, ret_seq);
+}
+
+extern "C" void stmgcn_lstm_fwd(void* stream_v, int dtype, const void* x,
+                                void* out, void* hseq_g, void* cseq_g,
+                                void* gates_g, const void** w_ih,
+                                const void** w_hh, const void** b_ih,
+                                const void** b_hh, int S, int Tst, int L,
+                                int cin, int ret_seq) {
+  RnnPtrs p;
+  for (int l = 0; l < L && l < MAX_LAYERS; ++l) {
+    p.w_ih[l] = w_ih[l]; p.w_hh[l] = w_hh[l];
+    p.b_ih[l] = b_ih[l]; p.b_hh[l] = b_hh[l];
+  }
+  hipStream_t stream = (hipStream_t)stream_v;
+  if (dtype == STM_BF16)
+    launch_fwd<__hip_bfloat16>(stream, x, out, hseq_g, cseq_g, gates_g, p, S,
+                               Tst, L, cin, ret_seq);
+  else if (dtype == STM_F16)
+    launch_fwd<__half>(stream, x, out, hseq_g, cseq_g, gates_g, p, S, Tst, L,
+                       cin, ret_seq);
+}
+
+extern "C" void stmgcn_mfma_probe(void* stream_v, const void* A, const void* B,
+                                  void* D) {
+  hipLaunchKernelGGL(mfma_probe_kernel, dim3(1), dim3(64), 0,
+                     (hipStream_t)stream_v, (const __hip_bfloat16*)A,
+                     (const __hip_bfloat16*)B, (float*)D);
+}
+
+// ===========================================================================
+// Fused LSTM backward, dgrad part (BPTT in-kernel) — SURVEY K10.
+//
+// Same workgroup geometry as forward. Layers walk top-down, timesteps in
+// reverse. Per step the pointwise phase turns (dh, dc, saved gates, saved
+// cell) into gate-preactivation grads dA — lane-local in the MFMA fragment
+// layout — then two MFMA GEMMs produce dh_{t-1} (recurrent carry, kept in
+// registers) and dx_t (written into the dh hand-off LDS slot for the layer
+// below, or to the dx output at layer 0). dA is also streamed to global in
+// natural (L,Tst,S_pad,4H) order; the weight gradients are then two plain
+// library GEMMs per layer on the host side (dW = dA^T @ [h_prev | x]),
+// which keeps this kernel lean (no long-lived dW accumulators).
+
+__device__ __forceinline__ int swzA(int s, int cbyte) {      // dA rows: 512 B
+  return s * 512 + (cbyte ^ ((s & 15) << 4));
+}
+
+template <typename T, bool CIN1>
+__global__ void __launch_bounds__(256, 1)
+lstm_bwd_kernel(const T* __restrict__ dout,     // (S,H) or (S,Tst,H)
+                const T* __restrict__ x,        // (S,Tst,Cin)
+                const float* __restrict__ cseq_g,
+                const T* __restrict__ gates_g,
+                RnnPtrs w,                      // w_ih/w_hh = TRANSPOSED (C|H, 4H)
+                T* __restrict__ dx,             // (S,Tst,Cin)
+                T* __restrict__ dA_g,           // (L,Tst,S_pad,4H)
+                int S, int Tst, int L, int ret_seq) {
+  using frag = typename Frag8<T>::type;
+  using elem = typename Frag8<T>::elem;
+  extern __shared__ char lds[];
+  const int s0 = blockIdx.x * SEQ_TILE;
+  const int wv = threadIdx.x >> 6;
+  const int lane = threadIdx.x & 63;
+  const int l16 = lane & 15;
+  const int lgrp = lane >> 4;
+  const long S_pad = (long)gridDim.x * SEQ_TILE;
+  const int hch = 16 * wv + l16;
+
+  char* dh_lds = lds;                         // [Tst][64][64] fwd-slot layout
+  char* dA_lds = lds + Tst * 8192;            // [64][256] swizzled (32 KiB)
+  float* red = (float*)(dA_lds + 64 * 512);   // [4][64] cross-wave scratch
+
+  for (int layer = L - 1; layer >= 0; --layer) {
+    const bool l0cin1 = CIN1 && layer == 0;
+    const T* WhhT = (const T*)w.w_hh[layer];  // (H, 4H)
+    const T* WihT = (const T*)w.w_ih[layer];  // (Cin_l, 4H)
+    float dh_rec[4][4], dc[4][4];
+    #pragma unroll
+    for (int m = 0; m < 4; ++m)
+      #pragma unroll
+      for (int r = 0; r < 4; ++r) { dh_rec[m][r] = 0.f; dc[m][r] = 0.f; }
+
+    for (int t = Tst - 1; t >= 0; --t) {
+      // WAR: previous step's reads of dA_lds must be complete
+      __syncthreads();
+
+      const long base = (long)layer * Tst + t;
+      const T* gp = gates_g + base * (S_pad * 4 * RNN_H)
+                    + (long)blockIdx.x * (SEQ_TILE * 4 * RNN_H) + (wv * 4) * 64 * 16;
+      const float* cp_t = cseq_g + base * (S_pad * RNN_H)
+                          + (long)blockIdx.x * (SEQ_TILE * RNN_H) + (wv * 4) * 64 * 4;
+      const float* cp_p = (t > 0) ? cp_t - (long)(S_pad * RNN_H) : nullptr;
+
+      float wih0[4];
+      if (l0cin1) {
+        #pragma unroll
+        for (int q = 0; q < 4; ++q) wih0[q] = toF<T>(WihT[q * 64 + hch]);
+      }
+      float dxpart[4][4];
+
+      #pragma unroll
+      for (int m = 0; m < 4; ++m) {
+        frag gf0 = *(((const frag*)(gp + (m * 64 + lane) * 16)) + 0); // i|f
+        frag gf1 = *(((const frag*)(gp + (m * 64 + lane) * 16)) + 1); // g|o
+        f32x4 ct = *(const f32x4*)(cp_t + (m * 64 + lane) * 4);
+        f32x4 cpv = cp_p ? *(const f32x4*)(cp_p + (m * 64 + lane) * 4)
+                         : f32x4{0.f, 0.f, 0.f, 0.f};
+        #pragma unroll
+        for (int r = 0; r < 4; ++r) {
+          const int row = 16 * m + 4 * lgrp + r;
+          float dh = dh_rec[m][r];
+          if (layer < L - 1) {
+            dh += toF<T>(*(const T*)&dh_lds[t * 8192 + lds_swz(row, hch * 2)]);
+          } else if (ret_seq) {
+            if (s0 + row < S) dh += toF<T>(dout[((long)(s0 + row) * Tst + t) * RNN_H + hch]);
+          } else if (t == Tst - 1) {
+            if (s0 + row < S) dh += toF<T>(dout[(long)(s0 + row) * RNN_H + hch]);
+          }
+          const float i_ = elemF(gf0[r]);
+          const float f_ = elemF(gf0[4 + r]);
+          const float g_ = elemF(gf1[r]);
+          const float o_ = elemF(gf1[4 + r]);
+          const float tc = tanhf(ct[r]);
+          float dcv = dc[m][r] + dh * o_ * (1.f - tc * tc);
+          const float dAo = dh * tc * o_ * (1.f - o_);
+          const float dAi = dcv * g_ * i_ * (1.f - i_);
+          const float dAf = dcv * cpv[r] * f_ * (1.f - f_);
+          const float dAg = dcv * i_ * (1.f - g_ * g_);
+          dc[m][r] = dcv * f_;
+          *(T*)&dA_lds[swzA(row, (0 * 64 + hch) * 2)] = fromF<T>(dAi);
+          *(T*)&dA_lds[swzA(row, (1 * 64 + hch) * 2)] = fromF<T>(dAf);
+          *(T*)&dA_lds[swzA(row, (2 * 64 + hch) * 2)] = fromF<T>(dAg);
+          *(T*)&dA_lds[swzA(row, (3 * 64 + hch) * 2)] = fromF<T>(dAo);
+          if (l0cin1)
+            dxpart[m][r] = dAi * wih0[0] + dAf * wih0[1] + dAg * wih0[2] + dAo * wih0[3];
+        }
+      }
+      __syncthreads();   // dA visible to all waves
+
+      // ---- stream dA to global (natural layout) for the host-side wgrad --
+      {
+        T* out_dA = dA_g + base * (S_pad * 4 * RNN_H) + (long)s0 * 4 * RNN_H;
+        for (int i = threadIdx.x; i < 64 * 32; i += 256) {
+          const int c8 = i & 31, sA = i >> 5;   // 32 x 16B pieces per row
+          *(frag*)&out_dA[(long)sA * 4 * RNN_H + c8 * 8] =
+              *(const frag*)&dA_lds[swzA(sA, c8 * 16)];
+        }
+      }
+
+      // ---- GEMM1: dh_prev = dA @ W_hh ------------------------------------
+      if (t > 0) {
+        f32x4 acc[4];
+        #pragma unroll
+        for (int m = 0; m < 4; ++m) acc[m] = f32x4{0.f, 0.f, 0.f, 0.f};
+        #pragma unroll 2
+        for (int kk = 0; kk < 8; ++kk) {
+          frag b = *(const frag*)&WhhT[hch * (4 * RNN_H) + kk * 32 + lgrp * 8];
+          #pragma unroll
+          for (int m = 0; m < 4; ++m) {
+            frag a = *(const frag*)&dA_lds[swzA(16 * m + l16, (kk * 32 + lgrp * 8) * 2)];
+            acc[m] = mfma16x16x32(a, b, acc[m]);
+          }
+        }
+        #pragma unroll
+        for (int m = 0; m < 4; ++m)
+          #pragma unroll
+          for (int r = 0; r < 4; ++r) dh_rec[m][r] = acc[m][r];
+      }
+
+      // ---- GEMM2: dx = dA @ W_ih -> dh_lds slot t / dx output ------------
+      if (!l0cin1) {
+        f32x4 acc[4];
+        #pragma unroll
+        for (int m = 0; m < 4; ++m) acc[m] = f32x4{0.f, 0.f, 0.f, 0.f};
+        #pragma unroll 2
+        for (int kk = 0; kk < 8; ++kk) {
+          frag b = *(const frag*)&WihT[hch * (4 * RNN_H) + kk * 32 + lgrp * 8];
+          #pragma unroll
+          for (int m = 0; m < 4; ++m) {
+            frag a = *(const frag*)&dA_lds[swzA(16 * m + l16, (kk * 32 + lgrp * 8) * 2)];
+            acc[m] = mfma16x16x32(a, b, acc[m]);
+          }
+        }
+        if (layer > 0) {
+          #pragma unroll
+          for (int m = 0; m < 4; ++m)
+            #pragma unroll
+            for (int r = 0; r < 4; ++r) {
+              const int row = 16 * m + 4 * lgrp + r;
+              *(T*)&dh_lds[t * 8192 + lds_swz(row, hch * 2)] = fromF<T>(acc[m][r]);
+            }
+        } else {
+          #pragma unroll
+          for (int m = 0; m < 4; ++m)
+            #pragma unroll
+            for (int r = 0; r < 4; ++r) {
+              const int row = 16 * m + 4 * lgrp + r;
+              if (s0 + row < S)
+                dx[((long)(s0 + row) * Tst + t) * RNN_H + hch] = fromF<T>(acc[m][r]);
+            }
+        }
+      } else {
+        // CIN1 l0: dx[s,t] = sum_g dA[s,g] W_ih[g,0]
+        #pragma unroll
+        for (int m = 0; m < 4; ++m)
+          #pragma unroll
+          for (int r = 0; r < 4; ++r) {
+            float v = dxpart[m][r];
+            #pragma unroll
+            for (int off = 1; off < 16; off <<= 1) v += __shfl_xor(v, off, 64);
+            if (l16 == 0) red[wv * 64 + 16 * m + 4 * lgrp + r] = v;
+          }
+        __syncthreads();
+        if (wv == 0) {
+          for (int sA = lane; sA < 64; sA += 64) {
+            const float v = red[sA] + red[64 + sA] + red[128 + sA] + red[192 + sA];
+            if (s0 + sA < S) dx[(long)(s0 + sA) * Tst + t] = fromF<T>(v);
+          }
+        }
+      }
+    }  // t loop
+  }  // layer loop
+}
+
+template <typename T>
+void launch_bwd(hipStream_t stream, const void* dout, const void* x,
+                const void* cseq_g, const void* gates_g, const RnnPtrs& w,
+                void* dx, void* dA_g, int S, int Tst, int L, int cin,
+                int ret_seq) {
+  const int nblk = (S + SEQ_TILE - 1) / SEQ_TILE;
+  const size_t lds_bytes = (size_t)Tst * 8192 + 64 * 512 + 4 * 64 * sizeof(float);
+  if (cin == 1)
+    hipLaunchKernelGGL((lstm_bwd_kernel<T, true>), dim3(nblk), dim3(256),
+                       lds_bytes, stream, (const T*)dout, (const T*)x,
+                       (const float*)cseq_g, (const T*)gates_g, w, (T*)dx,
+                       (T*)dA_g, S, Tst, L, ret_seq);
+  else
+    hipLaunchKernelGGL((lstm_bwd_kernel<T, false>), dim3(nblk), dim3(256),
+                       lds_bytes, stream, (const T*)dout, (const T*)x,
+                       (const float*)cseq_g, (const T*)gates_g, w, (T*)dx,
+                       (T*)dA_g, S, Tst, L, ret_seq);
+}
+
+extern "C" void stmgcn_lstm_bwd(void* stream_v, int dtype, const void* dout,
+                                const void* x, const void* cseq_g,
+                                const void* gates_g, const void** w_ihT,
+                                const void** w_hhT, void* dx, void* dA_g,
+                                int S, int Tst, int L, int cin, int ret_seq) {
+  RnnPtrs p;
+  for (int l = 0; l < L && l < MAX_LAYERS; ++l) {
+    p.w_ih[l] = w_ihT[l]; p.w_hh[l] = w_hhT[l];
+    p.b_ih[l] = nullptr; p.b_hh[l] = nullptr;
+  }
+  hipStream_t stream = (hipStream_t)stream_v;
+  if (dtype == STM_BF16)
+    launch_bwd<__hip_bfloat16>(stream, dout, x, cseq_g, gates_g, p, dx, dA_g,
+                               S, Tst, L, cin, ret_seq);
+  else if (dtype == STM_F16)
+    launch_bwd<__half>(stream, dout, x, cseq_g, gates_g, p, dx, dA_g, S, Tst,
+                       L, cin, ret_seq);
+}

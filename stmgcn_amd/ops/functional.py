@@ -83,6 +83,9 @@ def rnn_forward(cell: str, x: torch.Tensor, weights: List[torch.Tensor],
     if x.is_cuda and impl_mode() == "hip":
         from .hip_ops import FusedRNNFn
         return FusedRNNFn.apply(cell, x, h0, c0, return_sequences, *weights)
+    if x.is_cuda:  # stock-torch floor mode: the reference's nn.LSTM backend
+        from .hip_ops import _vf_rnn
+        return _vf_rnn(cell, x, h0, c0, return_sequences, list(weights))
     if cell == "lstm":
         return ref.lstm_forward(x, weights, h0, c0, return_sequences)
     if cell == "gru":

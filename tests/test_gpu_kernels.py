@@ -139,3 +139,73 @@ def test_train_step_gpu_bf16_finite():
         assert p.grad is None or torch.isfinite(p.grad.float()).all()
     opt.step()
     assert torch.isfinite(loss.float())
+
+
+def test_mfma_probe_layout():
+    """Validate the assumed 16x16x32 bf16 MFMA fragment maps on hardware.
+    ASYMMETRIC operands (transpose-detecting, guide §5.4 rule 16)."""
+    _C = require_hip()
+    torch.manual_seed(0)
+    A = (torch.randn(16, 32) * 0.5).bfloat16().cuda()
+    B = (torch.arange(32 * 16).float().reshape(32, 16) % 7 - 3).bfloat16().cuda() * 0.3
+    D = _C.mfma_probe(A, B)
+    ref_D = A.float().cpu() @ B.float().cpu()
+    torch.testing.assert_close(D.cpu(), ref_D, rtol=1e-2, atol=1e-2)
+
+
+def _lstm_oracle_weights(L, cin, H=64, seed=0):
+    torch.manual_seed(seed)
+    ws = []
+    for l in range(L):
+        in_l = cin if l == 0 else H
+        ws += [torch.randn(4 * H, in_l) * 0.2, torch.randn(4 * H, H) * 0.2,
+               torch.randn(4 * H) * 0.1, torch.randn(4 * H) * 0.1]
+    return ws
+
+
+@pytest.mark.parametrize("S,T,L,cin,ret_seq", [
+    (64, 8, 3, 1, False), (100, 8, 3, 1, False), (64, 5, 2, 1, True),
+    (64, 8, 2, 64, False), (128, 4, 1, 64, True)])
+def test_fused_lstm_forward_matches_oracle(S, T, L, cin, ret_seq):
+    from stmgcn_amd.ops.hip_ops import FusedLSTMFn
+    ws = _lstm_oracle_weights(L, cin)
+    x = torch.randn(S, T, cin)
+    out_ref = ref.lstm_forward(x, ws, torch.zeros(L, S, 64), torch.zeros(L, S, 64),
+                               ret_seq)
+    dev = torch.device("cuda")
+    ws_g = [w.bfloat16().to(dev) for w in ws]
+    out = FusedLSTMFn.apply(x.bfloat16().to(dev), ret_seq, False, *ws_g)
+    assert out.shape == out_ref.shape
+    err = (out.float().cpu() - out_ref).abs().max().item()
+    scale = out_ref.abs().max().item() + 1e-6
+    assert err / scale < 0.05, f"rel err {err/scale}"
+
+
+@pytest.mark.parametrize("S,T,L,cin,ret_seq", [
+    (64, 8, 3, 1, False), (100, 6, 2, 1, True), (64, 8, 2, 64, False)])
+def test_fused_lstm_backward_matches_oracle(S, T, L, cin, ret_seq):
+    from stmgcn_amd.ops.hip_ops import FusedLSTMFn
+    ws = _lstm_oracle_weights(L, cin)
+    x = torch.randn(S, T, cin)
+
+    x_ref = x.clone().requires_grad_(True)
+    ws_ref = [w.clone().requires_grad_(True) for w in ws]
+    out_ref = ref.lstm_forward(x_ref, ws_ref, torch.zeros(L, S, 64),
+                               torch.zeros(L, S, 64), ret_seq)
+    loss_ref = (out_ref.float() ** 2).sum()
+    loss_ref.backward()
+
+    dev = torch.device("cuda")
+    x_g = x.bfloat16().to(dev).requires_grad_(True)
+    ws_g = [w.bfloat16().to(dev).requires_grad_(True) for w in ws]
+    out = FusedLSTMFn.apply(x_g, ret_seq, True, *ws_g)
+    (out.float() ** 2).sum().backward()
+
+    def relerr(a, b):
+        return ((a.float().cpu() - b).abs().max() / (b.abs().max() + 1e-6)).item()
+
+    assert relerr(out.detach(), out_ref.detach()) < 0.05
+    assert relerr(x_g.grad, x_ref.grad) < 0.08, f"dx {relerr(x_g.grad, x_ref.grad)}"
+    for i, (wg, wr) in enumerate(zip(ws_g, ws_ref)):
+        e = relerr(wg.grad, wr.grad)
+        assert e < 0.08, f"weight {i} grad rel err {e}"
