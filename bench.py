@@ -46,6 +46,8 @@ def main():
     p.add_argument("--dtype", type=str, default=None, choices=sorted(DTYPES))
     p.add_argument("--profile-trace", type=str, default=None,
                    help="write a torch profiler trace to this path")
+    p.add_argument("--graph", type=str, default="auto", choices=["auto", "on", "off"],
+                   help="capture the whole train step in a hipGraph and replay it")
     args = p.parse_args()
 
     if args.impl:
@@ -83,13 +85,15 @@ def main():
     model = build_model(cfg).to(device=device, dtype=dtype)
     criterion = {"MSE": nn.MSELoss(), "MAE": nn.L1Loss(), "Huber": nn.SmoothL1Loss()}[cfg.loss]
     reducer = GradReducer(model) if world > 1 else None
-    opt = optim.Adam(model.parameters(), lr=cfg.lr, weight_decay=cfg.weight_decay)
+    want_graph = args.graph != "off" and use_gpu
+    opt = optim.Adam(model.parameters(), lr=cfg.lr, weight_decay=cfg.weight_decay,
+                     capturable=want_graph, foreach=True)
 
     def step():
         if reducer is not None:
             reducer.zero_grad()
         else:
-            opt.zero_grad(set_to_none=True)
+            opt.zero_grad(set_to_none=False)
         loss = criterion(model(x, adjs), y)
         loss.backward()
         if reducer is not None:
@@ -103,8 +107,37 @@ def main():
         if use_gpu:
             torch.cuda.synchronize()
 
-    for _ in range(args.warmup):
+    for _ in range(max(args.warmup, 2 if want_graph else 0)):
         step()
+    barrier_sync()
+
+    # ---- hipGraph capture of the whole training step ------------------------
+    # (forward + loss + backward (+ DP all-reduce) + Adam as ONE replayable
+    # graph — the model is launch-bound at this scale; see MI355X_MICROARCH
+    # "graph-replay-floor")
+    graph_ok = False
+    if want_graph:
+        try:
+            side = torch.cuda.Stream()
+            side.wait_stream(torch.cuda.current_stream())
+            with torch.cuda.stream(side):
+                step()
+            torch.cuda.current_stream().wait_stream(side)
+            torch.cuda.synchronize()
+            g = torch.cuda.CUDAGraph()
+            with torch.cuda.graph(g):
+                step()
+            torch.cuda.synchronize()
+            graph_ok = True
+        except Exception as e:
+            if args.graph == "on":
+                raise
+            print(f"# graph capture unavailable, eager fallback: {e}", flush=True)
+            graph_ok = False
+
+    run_step = (lambda: g.replay()) if graph_ok else step
+    for _ in range(args.warmup):
+        run_step()
     barrier_sync()
 
     prof_ctx = None
@@ -116,7 +149,7 @@ def main():
 
     t0 = time.perf_counter()
     for _ in range(args.steps):
-        step()
+        run_step()
     barrier_sync()
     elapsed = time.perf_counter() - t0
 
@@ -155,6 +188,7 @@ def main():
                 "lstm_hidden": cfg.lstm_hidden_dim, "gcn_hidden": cfg.gcn_hidden_dim,
                 "global_batch": B * n_gpus,
                 "parallelism": f"dp{n_gpus}",
+                "hipgraph": graph_ok,
                 "impl": os.environ.get("STMGCN_IMPL", "hip") if use_gpu else "torch-cpu",
             },
         }
