@@ -83,11 +83,24 @@ def main():
 
     # ---- model + optimizer --------------------------------------------------
     model = build_model(cfg).to(device=device, dtype=dtype)
-    criterion = {"MSE": nn.MSELoss(), "MAE": nn.L1Loss(), "Huber": nn.SmoothL1Loss()}[cfg.loss]
-    reducer = GradReducer(model) if world > 1 else None
     want_graph = args.graph != "off" and use_gpu
-    opt = optim.Adam(model.parameters(), lr=cfg.lr, weight_decay=cfg.weight_decay,
-                     capturable=want_graph, foreach=True)
+    hip_mode = use_gpu and os.environ.get("STMGCN_IMPL", "hip") == "hip"
+    if hip_mode and cfg.loss == "MSE":
+        from stmgcn_amd.ops import mse_loss as criterion
+    else:
+        criterion = {"MSE": nn.MSELoss(), "MAE": nn.L1Loss(),
+                     "Huber": nn.SmoothL1Loss()}[cfg.loss]
+    if hip_mode:
+        from stmgcn_amd.train import FusedAdam
+        # one flat arena; DP = one RCCL all-reduce over it (reduce())
+        opt = FusedAdam(model.parameters(), lr=cfg.lr,
+                        weight_decay=cfg.weight_decay)
+        reducer = None
+    else:
+        reducer = GradReducer(model) if world > 1 else None
+        opt = optim.Adam(model.parameters(), lr=cfg.lr,
+                         weight_decay=cfg.weight_decay,
+                         capturable=want_graph, foreach=True)
 
     def step():
         if reducer is not None:
@@ -98,6 +111,8 @@ def main():
         loss.backward()
         if reducer is not None:
             reducer.reduce()
+        elif hip_mode and world > 1:
+            opt.reduce()
         opt.step()
         return loss
 

@@ -225,10 +225,147 @@ at::Tensor mfma_probe(at::Tensor A, at::Tensor B) {
   return D;
 }
 
+// ---------------------------------------------------------------------------
+// pointwise.hip ops: K3 seqsum/permute, K4 gate, K7 head, K8 loss, K9 Adam
+
+extern "C" {
+void stmgcn_seqsum_permute(void*, int, const void*, void*, int, int, int, int);
+void stmgcn_seqsum_permute_bwd(void*, int, const void*, void*, int, int, int, int);
+void stmgcn_gate_fwd(void*, int, const void*, const void*, const void*,
+                     const void*, const void*, float*, float*, float*, void*,
+                     int, int, int, int);
+void stmgcn_gate_bwd(void*, int, const void*, const void*, const void*,
+                     const float*, const float*, const float*, float*, float*,
+                     float*, void*, void*, int, int, int, int);
+void stmgcn_head_fwd(void*, int, const void*, const void*, const void*,
+                     const void*, const void*, void*, void*, int, long);
+void stmgcn_head_bwd(void*, int, const void*, const void*, void*, int, long);
+void stmgcn_mse_fwd(void*, int, const void*, const void*, float*, void*, long);
+void stmgcn_mse_bwd(void*, int, const void*, const float*, void*, long);
+void stmgcn_adam(void*, int, float*, void*, const void*, float*, float*,
+                 float*, long);
+}
+
+at::Tensor seqsum_permute(at::Tensor obs) {
+  TORCH_CHECK(obs.is_cuda() && obs.dim() == 4 && obs.is_contiguous());
+  const int B = obs.size(0), Tst = obs.size(1), N = obs.size(2), C = obs.size(3);
+  auto out = at::empty({B, N, Tst}, obs.options());
+  stmgcn_seqsum_permute(stream(), dtype_code(obs), obs.data_ptr(),
+                        out.data_ptr(), B, Tst, N, C);
+  return out;
+}
+
+at::Tensor seqsum_permute_bwd(at::Tensor dxs, int64_t C) {
+  TORCH_CHECK(dxs.is_cuda() && dxs.dim() == 3);
+  dxs = dxs.contiguous();
+  const int B = dxs.size(0), N = dxs.size(1), Tst = dxs.size(2);
+  auto dobs = at::zeros({B, Tst, N, C}, dxs.options());
+  stmgcn_seqsum_permute_bwd(stream(), dtype_code(dxs), dxs.data_ptr(),
+                            dobs.data_ptr(), B, Tst, N, (int)C);
+  return dobs;
+}
+
+std::vector<at::Tensor> gate_fwd(at::Tensor obs, at::Tensor g, at::Tensor xs,
+                                 at::Tensor fcw, at::Tensor fcb) {
+  TORCH_CHECK(obs.is_cuda() && obs.dim() == 4 && obs.is_contiguous());
+  const int B = obs.size(0), Tst = obs.size(1), N = obs.size(2), C = obs.size(3);
+  TORCH_CHECK(Tst <= 16, "gate kernel serves T <= 16");
+  auto fopt = obs.options().dtype(at::kFloat);
+  auto z = at::empty({B, Tst}, fopt), u = at::empty({B, Tst}, fopt),
+       s = at::empty({B, Tst}, fopt);
+  auto out = at::empty_like(obs);
+  stmgcn_gate_fwd(stream(), dtype_code(obs), g.contiguous().data_ptr(),
+                  xs.contiguous().data_ptr(), fcw.contiguous().data_ptr(),
+                  fcb.contiguous().data_ptr(), obs.data_ptr(),
+                  z.data_ptr<float>(), u.data_ptr<float>(), s.data_ptr<float>(),
+                  out.data_ptr(), B, Tst, N, C);
+  return {out, z, u, s};
+}
+
+std::vector<at::Tensor> gate_bwd(at::Tensor dout, at::Tensor obs, at::Tensor fcw,
+                                 at::Tensor z, at::Tensor u, at::Tensor s) {
+  dout = dout.contiguous();
+  const int B = obs.size(0), Tst = obs.size(1), N = obs.size(2), C = obs.size(3);
+  auto fopt = obs.options().dtype(at::kFloat);
+  auto dz = at::empty({B, Tst}, fopt);
+  auto dw_part = at::empty({B, Tst, Tst}, fopt);
+  auto db_part = at::empty({B, Tst}, fopt);
+  auto dobs = at::empty_like(obs);
+  auto dg = at::empty({B, N, Tst}, obs.options());
+  stmgcn_gate_bwd(stream(), dtype_code(obs), dout.data_ptr(), obs.data_ptr(),
+                  fcw.contiguous().data_ptr(), z.data_ptr<float>(),
+                  u.data_ptr<float>(), s.data_ptr<float>(), dz.data_ptr<float>(),
+                  dw_part.data_ptr<float>(), db_part.data_ptr<float>(),
+                  dobs.data_ptr(), dg.data_ptr(), B, Tst, N, C);
+  return {dobs, dg, dw_part, db_part};
+}
+
+std::vector<at::Tensor> head_fwd(std::vector<at::Tensor> feats, at::Tensor w,
+                                 at::Tensor bias) {
+  auto f0 = feats[0].contiguous();
+  TORCH_CHECK(f0.is_cuda() && f0.dim() == 3 && feats.size() >= 1 && feats.size() <= 3);
+  const long BN = (long)f0.size(0) * f0.size(1);
+  const int G = f0.size(2);
+  TORCH_CHECK(G <= 64, "head kernel serves G <= 64");
+  auto y = at::empty({f0.size(0), f0.size(1), 1}, f0.options());
+  auto fsum = at::empty_like(f0);
+  TORCH_CHECK(bias.scalar_type() == f0.scalar_type());
+  stmgcn_head_fwd(stream(), dtype_code(f0), f0.data_ptr(),
+                  feats.size() > 1 ? feats[1].contiguous().data_ptr() : nullptr,
+                  feats.size() > 2 ? feats[2].contiguous().data_ptr() : nullptr,
+                  w.contiguous().data_ptr(), bias.contiguous().data_ptr(),
+                  y.data_ptr(), fsum.data_ptr(), G, BN);
+  return {y, fsum};
+}
+
+at::Tensor head_bwd(at::Tensor dy, at::Tensor w, int64_t G) {
+  dy = dy.contiguous();
+  const long BN = (long)dy.size(0) * dy.size(1);
+  auto dfeat = at::empty({dy.size(0), dy.size(1), G}, dy.options());
+  stmgcn_head_bwd(stream(), dtype_code(dy), dy.data_ptr(),
+                  w.contiguous().data_ptr(), dfeat.data_ptr(), (int)G, BN);
+  return dfeat;
+}
+
+std::vector<at::Tensor> mse_fwd(at::Tensor pred, at::Tensor tgt) {
+  pred = pred.contiguous();
+  auto loss = at::zeros({}, pred.options().dtype(at::kFloat));
+  auto diff = at::empty_like(pred);
+  stmgcn_mse_fwd(stream(), dtype_code(pred), pred.data_ptr(),
+                 tgt.contiguous().data_ptr(), loss.data_ptr<float>(),
+                 diff.data_ptr(), pred.numel());
+  return {loss, diff};
+}
+
+at::Tensor mse_bwd(at::Tensor diff, at::Tensor gscale) {
+  auto dpred = at::empty_like(diff);
+  stmgcn_mse_bwd(stream(), dtype_code(diff), diff.data_ptr(),
+                 gscale.contiguous().data_ptr<float>(), dpred.data_ptr(),
+                 diff.numel());
+  return dpred;
+}
+
+void adam_step(at::Tensor master, at::Tensor param, at::Tensor grad,
+               at::Tensor m, at::Tensor v, at::Tensor hyper) {
+  TORCH_CHECK(master.is_cuda() && master.scalar_type() == at::kFloat);
+  TORCH_CHECK(hyper.numel() == 9 && hyper.scalar_type() == at::kFloat);
+  stmgcn_adam(stream(), dtype_code(param), master.data_ptr<float>(),
+              param.data_ptr(), grad.data_ptr(), m.data_ptr<float>(),
+              v.data_ptr<float>(), hyper.data_ptr<float>(), master.numel());
+}
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("lstm_fwd", &lstm_fwd, "Fused multi-layer LSTM forward (persistent)");
   m.def("lstm_bwd", &lstm_bwd, "Fused LSTM dgrad (BPTT in-kernel)");
   m.def("mfma_probe", &mfma_probe, "16x16x32 bf16 MFMA fragment-layout probe");
+  m.def("seqsum_permute", &seqsum_permute, "K3: feature-sum + transpose");
+  m.def("seqsum_permute_bwd", &seqsum_permute_bwd);
+  m.def("gate_fwd", &gate_fwd, "K4: fused contextual gate forward");
+  m.def("gate_bwd", &gate_bwd, "K4: fused contextual gate backward");
+  m.def("head_fwd", &head_fwd, "K7: branch-sum + FC head forward");
+  m.def("head_bwd", &head_bwd, "K7: head backward (dfeat)");
+  m.def("mse_fwd", &mse_fwd, "K8: fused MSE loss forward");
+  m.def("mse_bwd", &mse_bwd, "K8: fused MSE grad");
+  m.def("adam_step", &adam_step, "K9: multi-tensor Adam over flat arena");
   m.def("cheb_apply", &cheb_apply,
         "Support stack S[b,n,k,c] = (T_k(G) x)[b,n,c] via in-kernel recurrence");
   m.def("cheb_combine", &cheb_combine,

@@ -124,7 +124,7 @@ class CG_LSTM(nn.Module):
                 return_sequences: bool = False) -> torch.Tensor:
         """obs_seq: (B, T, N, C) -> (B, N, H) (or (B, T, N, H) if sequences)."""
         B, T, N, C = obs_seq.shape
-        x_seq = obs_seq.sum(dim=-1).permute(0, 2, 1)          # (B,N,T), K3
+        x_seq = ops.seqsum_permute(obs_seq)                   # (B,N,T), K3
         g = self.gconv_temporal_feats(adj, x_seq)             # (B,N,T), K1+K2
         gated = ops.contextual_gate(obs_seq, g, self.fc.weight, self.fc.bias)  # K4
         flat = gated.permute(0, 2, 1, 3).reshape(B * N, T, C)  # node-major seqs

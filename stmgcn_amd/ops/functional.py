@@ -93,6 +93,22 @@ def rnn_forward(cell: str, x: torch.Tensor, weights: List[torch.Tensor],
     raise ValueError(f"unknown rnn cell {cell!r}")
 
 
+def seqsum_permute(obs_seq: torch.Tensor) -> torch.Tensor:
+    """K3: (B,T,N,C) -> (B,N,T) feature-sum + transpose (STMGCN.py:36,39)."""
+    if obs_seq.is_cuda and impl_mode() == "hip":
+        from .hip_ops import SeqsumPermuteFn
+        return SeqsumPermuteFn.apply(obs_seq)
+    return obs_seq.sum(dim=-1).permute(0, 2, 1)
+
+
+def mse_loss(pred: torch.Tensor, target: torch.Tensor) -> torch.Tensor:
+    """K8: fused MSE loss + grad on GPU; torch elsewhere."""
+    if pred.is_cuda and impl_mode() == "hip":
+        from .hip_ops import FusedMSELossFn
+        return FusedMSELossFn.apply(pred, target)
+    return torch.nn.functional.mse_loss(pred, target)
+
+
 def branch_fuse_head(branch_feats: List[torch.Tensor], fc_weight: torch.Tensor,
                      fc_bias: torch.Tensor) -> torch.Tensor:
     """Sum over M branches + FC head (reference STMGCN.py:116-118)."""

@@ -62,8 +62,8 @@ class ChebGconvFn(torch.autograd.Function):
         dz = dz.contiguous()
         B_, N, KC = feat.shape
         Cout = dz.shape[-1]
-        dW = (feat.reshape(-1, KC).T.to(torch.float32)
-              @ dz.reshape(-1, Cout).to(torch.float32)).to(W.dtype)
+        # bf16 GEMM accumulates fp32 inside rocBLAS; no fp32 cast kernels
+        dW = (feat.reshape(-1, KC).T @ dz.reshape(-1, Cout)).to(W.dtype)
         db = dz.sum(dim=(0, 1)).to(W.dtype) if ctx.has_b else None
         U = (dz @ W.to(dz.dtype).T).view(B_, N, csr.K_supports, ctx.cin).contiguous()
         dX = C.cheb_combine(U, csr.row_ptr_t, csr.col_idx_t, csr.vals_t,
@@ -72,7 +72,8 @@ class ChebGconvFn(torch.autograd.Function):
 
 
 def contextual_gate_hip(obs_seq, gconv_out, fc_weight, fc_bias):
-    # TODO(round1): fused cg_gate.hip kernel; interim torch path.
+    if obs_seq.shape[1] <= 16:
+        return GateFn.apply(obs_seq, gconv_out, fc_weight, fc_bias)
     return ref.contextual_gate(obs_seq, gconv_out, fc_weight, fc_bias)
 
 
@@ -139,14 +140,14 @@ class FusedLSTMFn(torch.autograd.Function):
             dA_l = dA[l].reshape(-1, 4 * H)                      # (Tst*S_pad, 4H)
             # h_{t-1}: shift hseq[l] right by one step
             hp = torch.cat([h_prev[0], hseq[l][:-1]], dim=0).reshape(-1, H)
-            dw_hh = (dA_l.t().float() @ hp.float()).to(w_hh[l].dtype)
+            dw_hh = dA_l.t() @ hp
             if l == 0:
                 xs = x.permute(1, 0, 2).reshape(Tst * S, cin)    # (Tst*S, C)
                 dA_x = dA[l][:, :S].reshape(-1, 4 * H)
-                dw_ih = (dA_x.t().float() @ xs.float()).to(w_ih[l].dtype)
+                dw_ih = dA_x.t() @ xs
             else:
                 xl = hseq[l - 1].reshape(-1, H)
-                dw_ih = (dA_l.t().float() @ xl.float()).to(w_ih[l].dtype)
+                dw_ih = dA_l.t() @ xl
             db = dA_l.sum(dim=0).to(w_ih[l].dtype)
             grads += [dw_ih, dw_hh, db, db.clone()]
         return (dx, None, None, *grads)
@@ -168,5 +169,91 @@ class FusedRNNFn:
 
 
 def branch_fuse_head_hip(branch_feats, fc_weight, fc_bias):
-    # TODO(round1): fused multi-graph-sum + FC head kernel.
+    if (fc_weight.shape[0] == 1 and fc_weight.shape[1] <= 64
+            and len(branch_feats) <= 3):
+        return HeadFn.apply(fc_weight, fc_bias, *branch_feats)
     return ref.branch_fuse_head(branch_feats, fc_weight, fc_bias)
+
+
+class SeqsumPermuteFn(torch.autograd.Function):
+    """K3: x_seq = obs.sum(-1).permute(0,2,1) as one kernel (reference
+    STMGCN.py:36,39); backward broadcasts over T,C."""
+
+    @staticmethod
+    def forward(ctx, obs):
+        C = require_hip()
+        ctx.C = obs.shape[-1]
+        return C.seqsum_permute(obs.contiguous())
+
+    @staticmethod
+    def backward(ctx, dxs):
+        C = require_hip()
+        return C.seqsum_permute_bwd(dxs, ctx.C)
+
+
+class GateFn(torch.autograd.Function):
+    """K4: fused contextual gate. The kernel folds the gradient of the
+    residual x_seq path (x_seq = sum_c obs) directly into dobs, so the xs
+    input gets no separate grad here; the gconv path (xs -> gconv -> g) flows
+    through dg and the autograd graph of xs as usual."""
+
+    @staticmethod
+    def forward(ctx, obs, g, fcw, fcb):
+        C = require_hip()
+        obs = obs.contiguous()
+        with torch.no_grad():
+            xs = C.seqsum_permute(obs)
+        out, z, u, s = C.gate_fwd(obs, g.contiguous(), xs, fcw.to(obs.dtype),
+                                  fcb.to(obs.dtype))
+        ctx.save_for_backward(obs, fcw, z, u, s)
+        return out
+
+    @staticmethod
+    def backward(ctx, dout):
+        C = require_hip()
+        obs, fcw, z, u, s = ctx.saved_tensors
+        dobs, dg, dw_part, db_part = C.gate_bwd(dout, obs, fcw.to(obs.dtype), z, u, s)
+        dw = dw_part.sum(dim=0).to(fcw.dtype)
+        db = db_part.sum(dim=0).to(fcw.dtype)
+        return dobs, dg, dw, db
+
+
+class HeadFn(torch.autograd.Function):
+    """K7: y = (sum_m feats_m) @ w^T + b fused (reference STMGCN.py:116-118)."""
+
+    @staticmethod
+    def forward(ctx, fc_weight, fc_bias, *feats):
+        C = require_hip()
+        y, fsum = C.head_fwd(list(feats), fc_weight.view(-1).to(feats[0].dtype),
+                             fc_bias.to(feats[0].dtype))
+        ctx.save_for_backward(fc_weight, fsum)
+        ctx.M = len(feats)
+        return y
+
+    @staticmethod
+    def backward(ctx, dy):
+        C = require_hip()
+        fc_weight, fsum = ctx.saved_tensors
+        G = fsum.shape[-1]
+        dfeat = C.head_bwd(dy, fc_weight.view(-1).to(dy.dtype), G)
+        dw = (dy.reshape(1, -1) @ fsum.reshape(-1, G)).to(fc_weight.dtype)
+        db = dy.sum().reshape(1).to(fc_weight.dtype)
+        return (dw, db) + (dfeat,) * ctx.M
+
+
+class FusedMSELossFn(torch.autograd.Function):
+    """K8: mean((pred-target)^2) with the grad fused (diff saved once)."""
+
+    @staticmethod
+    def forward(ctx, pred, target):
+        C = require_hip()
+        loss, diff = C.mse_fwd(pred, target)
+        ctx.save_for_backward(diff)
+        return loss
+
+    @staticmethod
+    def backward(ctx, gout):
+        C = require_hip()
+        (diff,) = ctx.saved_tensors
+        dpred = C.mse_bwd(diff, gout.detach().reshape(1).float().contiguous())
+        return dpred, None
