@@ -140,14 +140,22 @@ extern "C" void stmgcn_lstm_fwd(void* stream, int dtype, const void* x,
                                 void* gates_g, const void** w_ih,
                                 const void** w_hh, const void** b_ih,
                                 const void** b_hh, int S, int Tst, int L,
-                                int cin, int ret_seq);
+                                int cin, int ret_seq, int gru);
 extern "C" void stmgcn_lstm_bwd(void* stream, int dtype, const void* dout,
                                 const void* x, const void* cseq_g,
                                 const void* gates_g, const void** w_ihT,
                                 const void** w_hhT, void* dx, void* dA_g,
-                                int S, int Tst, int L, int cin, int ret_seq);
+                                int S, int Tst, int L, int cin, int ret_seq,
+                                int gru);
 extern "C" void stmgcn_mfma_probe(void* stream, const void* A, const void* B,
                                   void* D);
+extern "C" void stmgcn_lstm_wgrad(void* stream, int dtype, const void* dA,
+                                  const void* hseq, const void* x, float* dwih,
+                                  float* dwhh, float* db, long R, long S_pad,
+                                  int S, int Tst, int L, int cin);
+extern "C" void stmgcn_atb_wgrad(void* stream, int dtype, const void* A,
+                                 const void* B, float* C, float* db, long rows,
+                                 int M, int N);
 
 static constexpr int kSeqTile = 64;
 static constexpr int kH = 64;
@@ -156,7 +164,7 @@ std::vector<at::Tensor> lstm_fwd(at::Tensor x, std::vector<at::Tensor> w_ih,
                                  std::vector<at::Tensor> w_hh,
                                  std::vector<at::Tensor> b_ih,
                                  std::vector<at::Tensor> b_hh, bool ret_seq,
-                                 bool training) {
+                                 bool training, bool gru) {
   TORCH_CHECK(x.is_cuda() && x.dim() == 3 && x.is_contiguous());
   TORCH_CHECK(x.scalar_type() == at::kBFloat16 || x.scalar_type() == at::kHalf,
               "fused LSTM serves bf16/f16 (fp32 runs the torch path)");
@@ -189,7 +197,8 @@ std::vector<at::Tensor> lstm_fwd(at::Tensor x, std::vector<at::Tensor> w_ih,
     hp = hseq.data_ptr(); cp = cseq.data_ptr(); gp = gates.data_ptr();
   }
   stmgcn_lstm_fwd(stream(), dtype_code(x), x.data_ptr(), out.data_ptr(), hp,
-                  cp, gp, wi, wh, bi, bh, S, Tst, L, cin, ret_seq ? 1 : 0);
+                  cp, gp, wi, wh, bi, bh, S, Tst, L, cin, ret_seq ? 1 : 0,
+                  gru ? 1 : 0);
   if (!training) return {out};
   return {out, hseq, cseq, gates};
 }
@@ -197,7 +206,8 @@ std::vector<at::Tensor> lstm_fwd(at::Tensor x, std::vector<at::Tensor> w_ih,
 std::vector<at::Tensor> lstm_bwd(at::Tensor dout, at::Tensor x,
                                  at::Tensor cseq, at::Tensor gates,
                                  std::vector<at::Tensor> w_ihT,
-                                 std::vector<at::Tensor> w_hhT, bool ret_seq) {
+                                 std::vector<at::Tensor> w_hhT, bool ret_seq,
+                                 bool gru) {
   TORCH_CHECK(dout.is_cuda() && x.is_contiguous());
   dout = dout.contiguous();
   const int S = x.size(0), Tst = x.size(1), cin = x.size(2);
@@ -213,8 +223,45 @@ std::vector<at::Tensor> lstm_bwd(at::Tensor dout, at::Tensor x,
   }
   stmgcn_lstm_bwd(stream(), dtype_code(x), dout.data_ptr(), x.data_ptr(),
                   cseq.data_ptr(), gates.data_ptr(), wi, wh, dx.data_ptr(),
-                  dA.data_ptr(), S, Tst, L, cin, ret_seq ? 1 : 0);
+                  dA.data_ptr(), S, Tst, L, cin, ret_seq ? 1 : 0, gru ? 1 : 0);
   return {dx, dA};
+}
+
+// All-layer LSTM weight grads in ONE launch over the dA stream (wgrad.hip):
+// returns fp32 (dwih (L,4H,64), dwhh (L,4H,H), db (L,4H)); layer 0's live
+// dwih columns are [:, :cin].
+std::vector<at::Tensor> lstm_wgrad(at::Tensor dA, at::Tensor hseq, at::Tensor x) {
+  TORCH_CHECK(dA.is_cuda() && dA.dim() == 4 && dA.is_contiguous());
+  TORCH_CHECK(hseq.is_contiguous() && x.is_contiguous());
+  const int L = dA.size(0), Tst = dA.size(1);
+  const long S_pad = dA.size(2);
+  const long R = (long)Tst * S_pad;
+  const int S = x.size(0), cin = x.size(2);
+  TORCH_CHECK(cin == 1 || cin == kH);
+  auto fopt = dA.options().dtype(at::kFloat);
+  auto dwih = at::zeros({L, 4 * kH, kH}, fopt);
+  auto dwhh = at::zeros({L, 4 * kH, kH}, fopt);
+  auto db = at::zeros({L, 4 * kH}, fopt);
+  stmgcn_lstm_wgrad(stream(), dtype_code(dA), dA.data_ptr(), hseq.data_ptr(),
+                    x.data_ptr(), dwih.data_ptr<float>(), dwhh.data_ptr<float>(),
+                    db.data_ptr<float>(), R, S_pad, S, Tst, L, cin);
+  return {dwih, dwhh, db};
+}
+
+// C = A^T @ B (+ db = colsum(B)) for tall-skinny wgrads (M<=256, N<=64).
+std::vector<at::Tensor> atb_wgrad(at::Tensor A, at::Tensor B, bool want_db) {
+  TORCH_CHECK(A.is_cuda() && A.dim() == 2 && B.dim() == 2);
+  A = A.contiguous(); B = B.contiguous();
+  const long rows = A.size(0);
+  const int M = A.size(1), N = B.size(1);
+  TORCH_CHECK(B.size(0) == rows && M <= 256 && N <= 64);
+  auto fopt = A.options().dtype(at::kFloat);
+  auto C = at::zeros({M, N}, fopt);
+  auto db = want_db ? at::zeros({N}, fopt) : at::Tensor();
+  stmgcn_atb_wgrad(stream(), dtype_code(A), A.data_ptr(), B.data_ptr(),
+                   C.data_ptr<float>(), want_db ? db.data_ptr<float>() : nullptr,
+                   rows, M, N);
+  return want_db ? std::vector<at::Tensor>{C, db} : std::vector<at::Tensor>{C};
 }
 
 at::Tensor mfma_probe(at::Tensor A, at::Tensor B) {
@@ -356,6 +403,8 @@ void adam_step(at::Tensor master, at::Tensor param, at::Tensor grad,
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("lstm_fwd", &lstm_fwd, "Fused multi-layer LSTM forward (persistent)");
   m.def("lstm_bwd", &lstm_bwd, "Fused LSTM dgrad (BPTT in-kernel)");
+  m.def("lstm_wgrad", &lstm_wgrad, "All-layer LSTM weight grads, one launch");
+  m.def("atb_wgrad", &atb_wgrad, "C = A^T B reduction GEMM (+colsum(B))");
   m.def("mfma_probe", &mfma_probe, "16x16x32 bf16 MFMA fragment-layout probe");
   m.def("seqsum_permute", &seqsum_permute, "K3: feature-sum + transpose");
   m.def("seqsum_permute_bwd", &seqsum_permute_bwd);

@@ -85,14 +85,25 @@ extern "C" __global__ void mfma_probe_kernel(const __hip_bfloat16* A,
 }
 
 // ---------------------------------------------------------------------------
-// Fused LSTM forward.
+// Fused LSTM/GRU forward.
 //
 // Template: T = element type (bf16/f16), CIN1 = first-layer input_dim == 1
-// (base ST-MGCN; otherwise C_in == H, the stacked deep-variant blocks).
+// (base ST-MGCN; otherwise C_in == H, the stacked deep-variant blocks),
+// GRU = cell type (the deep variant, BASELINE configs[3]).
+//
+// GRU rides the 4-gate MFMA structure via host-side weight packing: the
+// 4 q-slots carry [r | z | n_input | n_hidden] with
+//   W_ih_packed = [Wr; Wz; Wn; 0],  W_hh_packed = [Ur; Uz; 0; Un],
+//   bias_packed = [bir+bhr; biz+bhz; bin; bhn]
+// so acc[2] = Wn x + bin and acc[3] = Un h + bhn arrive separately and the
+// pointwise phase computes n = tanh(acc2 + r*acc3), h' = (1-z) n + z h_prev.
+// The c_state register carry doubles as the GRU h_prev; the cseq training
+// save holds h_{t-1} (GRU) instead of c_t (LSTM). Everything else — LDS
+// staging, MFMA tiling, dA streaming, the batched wgrad kernel — is shared.
 //
 // LDS map (dynamic): hseq slots [T_steps][64][64] T-typed, swizzled;
 // then xbuf: CIN1 ? [T_steps][64] : [T_steps][64][64] (swizzled).
-template <typename T, bool CIN1>
+template <typename T, bool CIN1, bool GRU>
 __global__ void __launch_bounds__(256, 1)
 lstm_fwd_kernel(const T* __restrict__ x,    // (S, Tst, C_in)
                 T* __restrict__ out,        // (S, H) or (S, Tst, H)
@@ -217,7 +228,8 @@ lstm_fwd_kernel(const T* __restrict__ x,    // (S, Tst, C_in)
 
       // pointwise cell update in fragment layout (+ scalar-input term)
       T hval[4][4];     // [m][reg] this lane's h outputs (ch = hch)
-      T gsave[4][16];   // [m][i f g o][reg] post-activation gates
+      T gsave[4][16];   // [m][i f g o | r z n Bn][reg] post-activation gates
+      float csave[4][4];  // [m][reg] training save (LSTM c_t / GRU h_{t-1})
       #pragma unroll
       for (int m = 0; m < 4; ++m) {
         float xv[4];
@@ -235,16 +247,32 @@ lstm_fwd_kernel(const T* __restrict__ x,    // (S, Tst, C_in)
             gi += xv[r] * wih0[0]; gf += xv[r] * wih0[1];
             gg += xv[r] * wih0[2]; go += xv[r] * wih0[3];
           }
-          const float i_ = stm_sigmoid(gi), f_ = stm_sigmoid(gf);
-          const float g_ = tanhf(gg), o_ = stm_sigmoid(go);
-          const float c_ = f_ * c_state[m][r] + i_ * g_;
-          c_state[m][r] = c_;
-          const float h_ = o_ * tanhf(c_);
-          hval[m][r] = fromF<T>(h_);
-          gsave[m][0 * 4 + r] = fromF<T>(i_);
-          gsave[m][1 * 4 + r] = fromF<T>(f_);
-          gsave[m][2 * 4 + r] = fromF<T>(g_);
-          gsave[m][3 * 4 + r] = fromF<T>(o_);
+          if (GRU) {
+            // q-slots: gi = r-pre, gf = z-pre, gg = n_input, go = n_hidden
+            const float r_ = stm_sigmoid(gi), z_ = stm_sigmoid(gf);
+            const float n_ = tanhf(gg + r_ * go);
+            const float hprev = c_state[m][r];
+            csave[m][r] = hprev;                        // bwd needs h_{t-1}
+            const float h_ = (1.f - z_) * n_ + z_ * hprev;
+            c_state[m][r] = h_;
+            hval[m][r] = fromF<T>(h_);
+            gsave[m][0 * 4 + r] = fromF<T>(r_);
+            gsave[m][1 * 4 + r] = fromF<T>(z_);
+            gsave[m][2 * 4 + r] = fromF<T>(n_);
+            gsave[m][3 * 4 + r] = fromF<T>(go);          // Bn = Un h + bhn
+          } else {
+            const float i_ = stm_sigmoid(gi), f_ = stm_sigmoid(gf);
+            const float g_ = tanhf(gg), o_ = stm_sigmoid(go);
+            const float c_ = f_ * c_state[m][r] + i_ * g_;
+            c_state[m][r] = c_;
+            csave[m][r] = c_;
+            const float h_ = o_ * tanhf(c_);
+            hval[m][r] = fromF<T>(h_);
+            gsave[m][0 * 4 + r] = fromF<T>(i_);
+            gsave[m][1 * 4 + r] = fromF<T>(f_);
+            gsave[m][2 * 4 + r] = fromF<T>(g_);
+            gsave[m][3 * 4 + r] = fromF<T>(o_);
+          }
         }
       }
 
@@ -274,12 +302,13 @@ lstm_fwd_kernel(const T* __restrict__ x,    // (S, Tst, C_in)
           *(((frag*)(gp + (m * 64 + lane) * 16)) + 0) = *(frag*)&gsave[m][0],
           *(((frag*)(gp + (m * 64 + lane) * 16)) + 1) = *(frag*)&gsave[m][8];
         // cell: (L,Tst, S_pad*H) fp32 as [wave][m][lane][4]
+        // (LSTM: c_t; GRU: h_{t-1})
         float* cp = cseq_g + base * (S_pad * RNN_H)
                     + (long)blockIdx.x * (SEQ_TILE * RNN_H) + (wv * 4) * 64 * 4;
         #pragma unroll
       for (int m = 0; m < 4; ++m)
           *(f32x4*)(cp + (m * 64 + lane) * 4) =
-              f32x4{c_state[m][0], c_state[m][1], c_state[m][2], c_state[m][3]};
+              f32x4{csave[m][0], csave[m][1], csave[m][2], csave[m][3]};
         // hseq natural layout copy of slot t (also next layer's input source)
         T* hp = hseq_g + base * (S_pad * RNN_H) + (long)(s0)*RNN_H;
         char* slot = hseq + t * 8192;
@@ -314,18 +343,19 @@ lstm_fwd_kernel(const T* __restrict__ x,    // (S, Tst, C_in)
 template <typename T>
 void launch_fwd(hipStream_t stream, const void* x, void* out, void* hseq_g,
                 void* cseq_g, void* gates_g, const RnnPtrs& ptrs, int S,
-                int Tst, int L, int cin, int ret_seq) {
+                int Tst, int L, int cin, int ret_seq, int gru) {
   const int nblk = (S + SEQ_TILE - 1) / SEQ_TILE;
   const bool cin1 = (cin == 1);
   const size_t lds_bytes = (size_t)Tst * 8192 + (cin1 ? Tst * 64 * sizeof(T) : (size_t)Tst * 8192);
-  if (cin1)
-    hipLaunchKernelGGL((lstm_fwd_kernel<T, true>), dim3(nblk), dim3(256),
-                       lds_bytes, stream, (const T*)x, (T*)out, (T*)hseq_g,
-                       (float*)cseq_g, (T*)gates_g, ptrs, S, Tst, L, ret_seq);
-  else
-    hipLaunchKernelGGL((lstm_fwd_kernel<T, false>), dim3(nblk), dim3(256),
-                       lds_bytes, stream, (const T*)x, (T*)out, (T*)hseq_g,
-                       (float*)cseq_g, (T*)gates_g, ptrs, S, Tst, L, ret_seq);
+  auto go = [&](auto kern) {
+    hipLaunchKernelGGL(kern, dim3(nblk), dim3(256), lds_bytes, stream,
+                       (const T*)x, (T*)out, (T*)hseq_g, (float*)cseq_g,
+                       (T*)gates_g, ptrs, S, Tst, L, ret_seq);
+  };
+  if (cin1 && !gru) go(lstm_fwd_kernel<T, true, false>);
+  else if (cin1 && gru) go(lstm_fwd_kernel<T, true, true>);
+  else if (!gru) go(lstm_fwd_kernel<T, false, false>);
+  else go(lstm_fwd_kernel<T, false, true>);
 }
 
 extern "C" void stmgcn_lstm_fwd(void* stream_v, int dtype, const void* x,
@@ -333,7 +363,7 @@ extern "C" void stmgcn_lstm_fwd(void* stream_v, int dtype, const void* x,
                                 void* gates_g, const void** w_ih,
                                 const void** w_hh, const void** b_ih,
                                 const void** b_hh, int S, int Tst, int L,
-                                int cin, int ret_seq) {
+                                int cin, int ret_seq, int gru) {
   RnnPtrs p;
   for (int l = 0; l < L && l < MAX_LAYERS; ++l) {
     p.w_ih[l] = w_ih[l]; p.w_hh[l] = w_hh[l];
@@ -342,10 +372,10 @@ extern "C" void stmgcn_lstm_fwd(void* stream_v, int dtype, const void* x,
   hipStream_t stream = (hipStream_t)stream_v;
   if (dtype == STM_BF16)
     launch_fwd<__hip_bfloat16>(stream, x, out, hseq_g, cseq_g, gates_g, p, S,
-                               Tst, L, cin, ret_seq);
+                               Tst, L, cin, ret_seq, gru);
   else if (dtype == STM_F16)
     launch_fwd<__half>(stream, x, out, hseq_g, cseq_g, gates_g, p, S, Tst, L,
-                       cin, ret_seq);
+                       cin, ret_seq, gru);
 }
 
 extern "C" void stmgcn_mfma_probe(void* stream_v, const void* A, const void* B,
@@ -372,7 +402,12 @@ __device__ __forceinline__ int swzA(int s, int cbyte) {      // dA rows: 512 B
   return s * 512 + (cbyte ^ ((s & 15) << 4));
 }
 
-template <typename T, bool CIN1>
+// GRU backward (same packed-slot scheme as forward — see lstm_fwd_kernel):
+// saved gates = [r, z, n, Bn], cseq = h_{t-1}; emits packed
+// dA = [dr_pre, dz_pre, dn_pre, dBn] so the dgrad GEMMs against the packed
+// transposed weights and the batched wgrad kernel run unchanged. The direct
+// dh_{t-1} += dh_t * z term rides the dc[][] register carry.
+template <typename T, bool CIN1, bool GRU>
 __global__ void __launch_bounds__(256, 1)
 lstm_bwd_kernel(const T* __restrict__ dout,     // (S,H) or (S,Tst,H)
                 const T* __restrict__ x,        // (S,Tst,Cin)
@@ -443,17 +478,34 @@ lstm_bwd_kernel(const T* __restrict__ dout,     // (S,H) or (S,Tst,H)
           } else if (t == Tst - 1) {
             if (s0 + row < S) dh += toF<T>(dout[(long)(s0 + row) * RNN_H + hch]);
           }
-          const float i_ = elemF(gf0[r]);
-          const float f_ = elemF(gf0[4 + r]);
-          const float g_ = elemF(gf1[r]);
-          const float o_ = elemF(gf1[4 + r]);
-          const float tc = tanhf(ct[r]);
-          float dcv = dc[m][r] + dh * o_ * (1.f - tc * tc);
-          const float dAo = dh * tc * o_ * (1.f - o_);
-          const float dAi = dcv * g_ * i_ * (1.f - i_);
-          const float dAf = dcv * cpv[r] * f_ * (1.f - f_);
-          const float dAg = dcv * i_ * (1.f - g_ * g_);
-          dc[m][r] = dcv * f_;
+          float dAi, dAf, dAg, dAo;
+          if (GRU) {
+            // gf0 = [r, z], gf1 = [n, Bn]; ct = h_{t-1} (fwd csave)
+            const float r_ = elemF(gf0[r]);
+            const float z_ = elemF(gf0[4 + r]);
+            const float n_ = elemF(gf1[r]);
+            const float Bn = elemF(gf1[4 + r]);
+            const float hprev = ct[r];
+            dh += dc[m][r];                       // direct dh_t * z carry
+            const float dz_pre = dh * (hprev - n_) * z_ * (1.f - z_);
+            const float dn_pre = dh * (1.f - z_) * (1.f - n_ * n_);
+            const float dBn = dn_pre * r_;
+            const float dr_pre = dn_pre * Bn * r_ * (1.f - r_);
+            dc[m][r] = dh * z_;
+            dAi = dr_pre; dAf = dz_pre; dAg = dn_pre; dAo = dBn;
+          } else {
+            const float i_ = elemF(gf0[r]);
+            const float f_ = elemF(gf0[4 + r]);
+            const float g_ = elemF(gf1[r]);
+            const float o_ = elemF(gf1[4 + r]);
+            const float tc = tanhf(ct[r]);
+            float dcv = dc[m][r] + dh * o_ * (1.f - tc * tc);
+            dAo = dh * tc * o_ * (1.f - o_);
+            dAi = dcv * g_ * i_ * (1.f - i_);
+            dAf = dcv * cpv[r] * f_ * (1.f - f_);
+            dAg = dcv * i_ * (1.f - g_ * g_);
+            dc[m][r] = dcv * f_;
+          }
           *(T*)&dA_lds[swzA(row, (0 * 64 + hch) * 2)] = fromF<T>(dAi);
           *(T*)&dA_lds[swzA(row, (1 * 64 + hch) * 2)] = fromF<T>(dAf);
           *(T*)&dA_lds[swzA(row, (2 * 64 + hch) * 2)] = fromF<T>(dAg);
@@ -553,26 +605,27 @@ template <typename T>
 void launch_bwd(hipStream_t stream, const void* dout, const void* x,
                 const void* cseq_g, const void* gates_g, const RnnPtrs& w,
                 void* dx, void* dA_g, int S, int Tst, int L, int cin,
-                int ret_seq) {
+                int ret_seq, int gru) {
   const int nblk = (S + SEQ_TILE - 1) / SEQ_TILE;
   const size_t lds_bytes = (size_t)Tst * 8192 + 64 * 512 + 4 * 64 * sizeof(float);
-  if (cin == 1)
-    hipLaunchKernelGGL((lstm_bwd_kernel<T, true>), dim3(nblk), dim3(256),
-                       lds_bytes, stream, (const T*)dout, (const T*)x,
-                       (const float*)cseq_g, (const T*)gates_g, w, (T*)dx,
-                       (T*)dA_g, S, Tst, L, ret_seq);
-  else
-    hipLaunchKernelGGL((lstm_bwd_kernel<T, false>), dim3(nblk), dim3(256),
-                       lds_bytes, stream, (const T*)dout, (const T*)x,
-                       (const float*)cseq_g, (const T*)gates_g, w, (T*)dx,
-                       (T*)dA_g, S, Tst, L, ret_seq);
+  auto go = [&](auto kern) {
+    hipLaunchKernelGGL(kern, dim3(nblk), dim3(256), lds_bytes, stream,
+                       (const T*)dout, (const T*)x, (const float*)cseq_g,
+                       (const T*)gates_g, w, (T*)dx, (T*)dA_g, S, Tst, L,
+                       ret_seq);
+  };
+  if (cin == 1 && !gru) go(lstm_bwd_kernel<T, true, false>);
+  else if (cin == 1 && gru) go(lstm_bwd_kernel<T, true, true>);
+  else if (!gru) go(lstm_bwd_kernel<T, false, false>);
+  else go(lstm_bwd_kernel<T, false, true>);
 }
 
 extern "C" void stmgcn_lstm_bwd(void* stream_v, int dtype, const void* dout,
                                 const void* x, const void* cseq_g,
                                 const void* gates_g, const void** w_ihT,
                                 const void** w_hhT, void* dx, void* dA_g,
-                                int S, int Tst, int L, int cin, int ret_seq) {
+                                int S, int Tst, int L, int cin, int ret_seq,
+                                int gru) {
   RnnPtrs p;
   for (int l = 0; l < L && l < MAX_LAYERS; ++l) {
     p.w_ih[l] = w_ihT[l]; p.w_hh[l] = w_hhT[l];
@@ -581,8 +634,8 @@ extern "C" void stmgcn_lstm_bwd(void* stream_v, int dtype, const void* dout,
   hipStream_t stream = (hipStream_t)stream_v;
   if (dtype == STM_BF16)
     launch_bwd<__hip_bfloat16>(stream, dout, x, cseq_g, gates_g, p, dx, dA_g,
-                               S, Tst, L, cin, ret_seq);
+                               S, Tst, L, cin, ret_seq, gru);
   else if (dtype == STM_F16)
     launch_bwd<__half>(stream, dout, x, cseq_g, gates_g, p, dx, dA_g, S, Tst,
-                       L, cin, ret_seq);
+                       L, cin, ret_seq, gru);
 }

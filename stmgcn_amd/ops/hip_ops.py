@@ -62,9 +62,16 @@ class ChebGconvFn(torch.autograd.Function):
         dz = dz.contiguous()
         B_, N, KC = feat.shape
         Cout = dz.shape[-1]
-        # bf16 GEMM accumulates fp32 inside rocBLAS; no fp32 cast kernels
-        dW = (feat.reshape(-1, KC).T @ dz.reshape(-1, Cout)).to(W.dtype)
-        db = dz.sum(dim=(0, 1)).to(W.dtype) if ctx.has_b else None
+        # tall-skinny reduction GEMM kernel (wgrad.hip): dW = feat^T dZ and
+        # db = colsum(dZ) in one launch, fp32 accumulate. (bf16/f16 only —
+        # the MFMA tile is 16x16x32_bf16/f16; fp32 parity runs rocBLAS.)
+        if dz.dtype in (torch.bfloat16, torch.float16) and KC <= 256 and Cout <= 64:
+            outs = C.atb_wgrad(feat.reshape(-1, KC), dz.reshape(-1, Cout), ctx.has_b)
+            dW = outs[0].to(W.dtype)
+            db = outs[1].to(W.dtype) if ctx.has_b else None
+        else:
+            dW = (feat.reshape(-1, KC).T @ dz.reshape(-1, Cout)).to(W.dtype)
+            db = dz.sum(dim=(0, 1)).to(W.dtype) if ctx.has_b else None
         U = (dz @ W.to(dz.dtype).T).view(B_, N, csr.K_supports, ctx.cin).contiguous()
         dX = C.cheb_combine(U, csr.row_ptr_t, csr.col_idx_t, csr.vals_t,
                             csr.kind == "single")
@@ -91,26 +98,55 @@ def _vf_rnn(cell, x, h0, c0, return_sequences, weights):
 
 
 class FusedLSTMFn(torch.autograd.Function):
-    """Persistent fused multi-layer LSTM: one kernel for
-    all layers x timesteps (forward), one dgrad kernel for the BPTT; weight
-    gradients are two plain library GEMMs per layer over the streamed
-    gate-preactivation grads dA (dW = dA^T @ [h_prev | x]).
+    """Persistent fused multi-layer LSTM/GRU: one kernel
+    for all layers x timesteps (forward), one dgrad kernel for the BPTT, one
+    batched reduction-GEMM kernel (wgrad.hip) for ALL weight gradients.
+
+    GRU (the deep variant, BASELINE configs[3]) rides the same 4-slot MFMA
+    kernels through weight packing — q-slots [r | z | n_input | n_hidden]:
+      W_ih_packed = [Wr; Wz; Wn; 0],  W_hh_packed = [Ur; Uz; 0; Un]
+      b_ih_packed = [bir; biz; bin; 0], b_hh_packed = [bhr; bhz; 0; bhn]
+    and the packed dA stream [dr|dz|dn|dBn] flows through the shared dgrad
+    GEMMs and wgrad kernel; grads are un-packed by slicing below.
 
     Constraints (checked): bf16/f16, H == 64, C_in in {1, 64}, T <= 16,
     L <= 8, zero initial states (the reference zero-inits every forward —
     STMGCN.py:93-98,109, quirk 8)."""
 
     @staticmethod
-    def forward(ctx, x, ret_seq, training, *weights):
+    def forward(ctx, x, cell, ret_seq, training, *weights):
         C = require_hip()
         L = len(weights) // 4
-        w_ih = [weights[4 * l + 0].contiguous() for l in range(L)]
-        w_hh = [weights[4 * l + 1].contiguous() for l in range(L)]
-        b_ih = [weights[4 * l + 2].float().contiguous() for l in range(L)]
-        b_hh = [weights[4 * l + 3].float().contiguous() for l in range(L)]
-        outs = C.lstm_fwd(x.contiguous(), w_ih, w_hh, b_ih, b_hh, ret_seq, training)
+        gru = cell == "gru"
+        if gru:
+            H = weights[1].shape[1]
+            w_ih, w_hh, b_ih, b_hh = [], [], [], []
+            for l in range(L):
+                wi, wh, bi, bh = weights[4 * l:4 * l + 4]
+                wp = wi.new_zeros(4 * H, wi.shape[1])
+                wp[:3 * H] = wi
+                hp = wh.new_zeros(4 * H, H)
+                hp[:2 * H] = wh[:2 * H]
+                hp[3 * H:] = wh[2 * H:]
+                bip = torch.zeros(4 * H, dtype=torch.float32, device=wi.device)
+                bhp = torch.zeros_like(bip)
+                bip[:3 * H] = bi.float()
+                bhp[:2 * H] = bh[:2 * H].float()
+                bhp[3 * H:] = bh[2 * H:].float()
+                w_ih.append(wp.contiguous())
+                w_hh.append(hp.contiguous())
+                b_ih.append(bip)
+                b_hh.append(bhp)
+        else:
+            w_ih = [weights[4 * l + 0].contiguous() for l in range(L)]
+            w_hh = [weights[4 * l + 1].contiguous() for l in range(L)]
+            b_ih = [weights[4 * l + 2].float().contiguous() for l in range(L)]
+            b_hh = [weights[4 * l + 3].float().contiguous() for l in range(L)]
+        outs = C.lstm_fwd(x.contiguous(), w_ih, w_hh, b_ih, b_hh, ret_seq,
+                          training, gru)
         ctx.ret_seq = ret_seq
         ctx.L = L
+        ctx.gru = gru
         if training:
             out, hseq, cseq, gates = outs
             ctx.save_for_backward(x, cseq, gates, hseq, *w_ih, *w_hh)
@@ -123,48 +159,52 @@ class FusedLSTMFn(torch.autograd.Function):
         C = require_hip()
         L = ctx.L
         x, cseq, gates, hseq = ctx.saved_tensors[:4]
-        w_ih = ctx.saved_tensors[4:4 + L]
+        w_ih = ctx.saved_tensors[4:4 + L]        # packed for GRU
         w_hh = ctx.saved_tensors[4 + L:4 + 2 * L]
         w_ihT = [w.t().contiguous() for w in w_ih]
         w_hhT = [w.t().contiguous() for w in w_hh]
         x = x.contiguous()
-        dx, dA = C.lstm_bwd(dout, x, cseq, gates, w_ihT, w_hhT, ctx.ret_seq)
+        dx, dA = C.lstm_bwd(dout, x, cseq, gates, w_ihT, w_hhT, ctx.ret_seq,
+                            ctx.gru)
         S, Tst, cin = x.shape
-        # ---- weight grads: plain GEMMs over the dA stream (rocBLAS) --------
-        # dA: (L, Tst, S_pad, 4H); hseq: (L, Tst, S_pad, H)
-        S_pad = dA.shape[2]
-        H = hseq.shape[-1]
+        # ---- weight grads: ONE batched reduction-GEMM launch for all layers
+        # (wgrad.hip; replaces 5 hipBLASLt GEMMs + 3 cat + 3 colsum per call
+        # that measured 47.8% of step time — profiles/r01_*)
+        dwih, dwhh, dbf = C.lstm_wgrad(dA, hseq, x)
         grads = []
-        h_prev = torch.zeros_like(hseq[:, :1])
+        H = 64
         for l in range(L):
-            dA_l = dA[l].reshape(-1, 4 * H)                      # (Tst*S_pad, 4H)
-            # h_{t-1}: shift hseq[l] right by one step
-            hp = torch.cat([h_prev[0], hseq[l][:-1]], dim=0).reshape(-1, H)
-            dw_hh = dA_l.t() @ hp
-            if l == 0:
-                xs = x.permute(1, 0, 2).reshape(Tst * S, cin)    # (Tst*S, C)
-                dA_x = dA[l][:, :S].reshape(-1, 4 * H)
-                dw_ih = dA_x.t() @ xs
+            dt = w_ih[l].dtype
+            cin_l = cin if l == 0 else H
+            if ctx.gru:
+                dw_ih = dwih[l, :3 * H, :cin_l].to(dt)
+                dw_hh = torch.cat([dwhh[l, :2 * H], dwhh[l, 3 * H:]]).to(dt)
+                db_ih = dbf[l, :3 * H].to(dt)
+                db_hh = torch.cat([dbf[l, :2 * H], dbf[l, 3 * H:]]).to(dt)
             else:
-                xl = hseq[l - 1].reshape(-1, H)
-                dw_ih = dA_l.t() @ xl
-            db = dA_l.sum(dim=0).to(w_ih[l].dtype)
-            grads += [dw_ih, dw_hh, db, db.clone()]
-        return (dx, None, None, *grads)
+                dw_ih = dwih[l, :, :cin_l].to(dt)
+                dw_hh = dwhh[l].to(dt)
+                db_ih = dbf[l].to(dt)
+                db_hh = db_ih.clone()
+            grads += [dw_ih, dw_hh, db_ih, db_hh]
+        return (dx, None, None, None, *grads)
 
 
 class FusedRNNFn:
-    """Dispatch: LSTM -> fused HIP kernels (bf16/f16); GRU and fp32 -> torch
-    native fused RNN (interim; GRU HIP kernels land with the deep variant)."""
+    """Dispatch: LSTM/GRU -> fused HIP kernels (bf16/f16, H=64);
+    fp32 or off-shape -> torch native fused RNN (the stock floor)."""
 
     @staticmethod
     def apply(cell, x, h0, c0, return_sequences, *weights):
-        if (cell == "lstm" and x.dtype in (torch.bfloat16, torch.float16)
-                and x.shape[-1] in (1, 64) and weights[1].shape[1] == 64
+        gmul = 4 if cell == "lstm" else 3
+        if (cell in ("lstm", "gru")
+                and x.dtype in (torch.bfloat16, torch.float16)
+                and x.shape[-1] in (1, 64)
+                and weights[1].shape == (gmul * 64, 64)
                 and x.shape[1] <= 16):
             training = torch.is_grad_enabled() and (
                 x.requires_grad or any(w.requires_grad for w in weights))
-            return FusedLSTMFn.apply(x, return_sequences, training, *weights)
+            return FusedLSTMFn.apply(x, cell, return_sequences, training, *weights)
         return _vf_rnn(cell, x, h0, c0, return_sequences, weights)
 
 
