@@ -271,7 +271,12 @@ extern "C" void stmgcn_lstm_wgrad(void* stream_v, int dtype, const void* dA,
                                   const void* hseq, const void* x, float* dwih,
                                   float* dwhh, float* db, long R, long S_pad,
                                   int S, int Tst, int L, int cin) {
-  const long target = 512 / (L > 0 ? L : 1);
+  static long env_chunks = -1;   // STMGCN_WGRAD_CHUNKS: perf experiments
+  if (env_chunks < 0) {
+    const char* e = getenv("STMGCN_WGRAD_CHUNKS");
+    env_chunks = e ? atol(e) : 0;
+  }
+  const long target = env_chunks > 0 ? env_chunks : 512 / (L > 0 ? L : 1);
   long nchunks = (R + 1023) / 1024;
   if (nchunks > target) nchunks = target;
   if (nchunks < 1) nchunks = 1;
