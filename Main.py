@@ -108,10 +108,24 @@ def main():
     # ---- model (L4) ----
     model = build_model(cfg).to(device=device, dtype=dtype)
 
-    loss = {"MSE": nn.MSELoss(), "MAE": nn.L1Loss(), "Huber": nn.SmoothL1Loss()}[cfg.loss]
-    reducer = GradReducer(model) if world > 1 else None
+    # MI355X path: fused MSE loss + flat-arena FusedAdam (one Adam launch,
+    # one flat RCCL all-reduce for DP); torch Adam + GradReducer otherwise.
+    from stmgcn_amd import ops as _ops
+    hip_path = (device.type == "cuda" and _ops.hip_available()
+                and dtype in (torch.bfloat16, torch.float16)
+                and os.environ.get("STMGCN_IMPL", "hip") == "hip")
+    if hip_path and cfg.loss == "MSE":
+        from stmgcn_amd.ops import mse_loss as loss
+    else:
+        loss = {"MSE": nn.MSELoss(), "MAE": nn.L1Loss(), "Huber": nn.SmoothL1Loss()}[cfg.loss]
+    if hip_path:
+        from stmgcn_amd.train import FusedAdam as opt_cls
+        reducer = None
+    else:
+        opt_cls = optim.Adam
+        reducer = GradReducer(model) if world > 1 else None
 
-    trainer = ModelTrainer(model=model, loss=loss, optimizer=optim.Adam,
+    trainer = ModelTrainer(model=model, loss=loss, optimizer=opt_cls,
                            lr=cfg.lr, wd=cfg.weight_decay, n_epochs=cfg.n_epochs,
                            grad_reducer=reducer, rank=rank, world_size=world,
                            metrics_path=args.metrics)

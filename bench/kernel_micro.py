@@ -10,6 +10,9 @@ import argparse
 import json
 import os
 
+import sys
+sys.path.insert(0, __import__("os").path.dirname(__import__("os").path.dirname(__import__("os").path.abspath(__file__))))
+
 import torch
 
 

@@ -176,8 +176,8 @@ std::vector<at::Tensor> lstm_fwd(at::Tensor x, std::vector<at::Tensor> w_ih,
     TORCH_CHECK(w_hh[l].size(0) == 4 * kH && w_hh[l].size(1) == kH,
                 "hidden dim must be 64");
     TORCH_CHECK(w_ih[l].is_contiguous() && w_hh[l].is_contiguous());
-    TORCH_CHECK(b_ih[l].scalar_type() == at::kFloat &&
-                b_hh[l].scalar_type() == at::kFloat);
+    TORCH_CHECK(b_ih[l].scalar_type() == x.scalar_type() &&
+                b_hh[l].scalar_type() == x.scalar_type());
   }
   const long nblk = (S + kSeqTile - 1) / kSeqTile;
   const long S_pad = nblk * kSeqTile;

@@ -37,8 +37,8 @@ typedef __attribute__((ext_vector_type(4))) float f32x4;
 struct RnnPtrs {
   const void* w_ih[MAX_LAYERS];   // (G*H, C_l) row-major, model dtype
   const void* w_hh[MAX_LAYERS];   // (G*H, H)
-  const void* b_ih[MAX_LAYERS];   // (G*H,) fp32
-  const void* b_hh[MAX_LAYERS];   // (G*H,) fp32
+  const void* b_ih[MAX_LAYERS];   // (G*H,) model dtype
+  const void* b_hh[MAX_LAYERS];   // (G*H,) model dtype
 };
 
 // ---------------------------------------------------------------------------
@@ -153,8 +153,8 @@ lstm_fwd_kernel(const T* __restrict__ x,    // (S, Tst, C_in)
     const int cin = (layer == 0) ? (CIN1 ? 1 : RNN_H) : RNN_H;
     const T* Whh = (const T*)ptrs.w_hh[layer];
     const T* Wih = (const T*)ptrs.w_ih[layer];
-    const float* bih = (const float*)ptrs.b_ih[layer];
-    const float* bhh = (const float*)ptrs.b_hh[layer];
+    const T* bih = (const T*)ptrs.b_ih[layer];
+    const T* bhh = (const T*)ptrs.b_hh[layer];
 
     // B-fragments for the recurrent GEMM: b_hh[q][kk]; W row g = q*64 + hch,
     // cols kk*32 + lgrp*8 .. +8 -> contiguous 16B in the (4H, H) weight.
@@ -178,7 +178,7 @@ lstm_fwd_kernel(const T* __restrict__ x,    // (S, Tst, C_in)
     #pragma unroll
       for (int q = 0; q < 4; ++q) {
       const int g = q * 64 + hch;
-      bias[q] = bih[g] + bhh[g];
+      bias[q] = toF<T>(bih[g]) + toF<T>(bhh[g]);
       if (CIN1 && layer == 0) wih0[q] = toF<T>(Wih[g]);
     }
 

@@ -128,11 +128,11 @@ class FusedLSTMFn(torch.autograd.Function):
                 hp = wh.new_zeros(4 * H, H)
                 hp[:2 * H] = wh[:2 * H]
                 hp[3 * H:] = wh[2 * H:]
-                bip = torch.zeros(4 * H, dtype=torch.float32, device=wi.device)
+                bip = wi.new_zeros(4 * H)
                 bhp = torch.zeros_like(bip)
-                bip[:3 * H] = bi.float()
-                bhp[:2 * H] = bh[:2 * H].float()
-                bhp[3 * H:] = bh[2 * H:].float()
+                bip[:3 * H] = bi
+                bhp[:2 * H] = bh[:2 * H]
+                bhp[3 * H:] = bh[2 * H:]
                 w_ih.append(wp.contiguous())
                 w_hh.append(hp.contiguous())
                 b_ih.append(bip)
@@ -140,8 +140,8 @@ class FusedLSTMFn(torch.autograd.Function):
         else:
             w_ih = [weights[4 * l + 0].contiguous() for l in range(L)]
             w_hh = [weights[4 * l + 1].contiguous() for l in range(L)]
-            b_ih = [weights[4 * l + 2].float().contiguous() for l in range(L)]
-            b_hh = [weights[4 * l + 3].float().contiguous() for l in range(L)]
+            b_ih = [weights[4 * l + 2].contiguous() for l in range(L)]
+            b_hh = [weights[4 * l + 3].contiguous() for l in range(L)]
         outs = C.lstm_fwd(x.contiguous(), w_ih, w_hh, b_ih, b_hh, ret_seq,
                           training, gru)
         ctx.ret_seq = ret_seq

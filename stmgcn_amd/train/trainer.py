@@ -107,11 +107,15 @@ class ModelTrainer:
                         if mode == "train":
                             if self.grad_reducer is not None:
                                 self.grad_reducer.zero_grad()
+                            elif hasattr(self.optimizer, "reduce"):
+                                self.optimizer.zero_grad()   # FusedAdam arena
                             else:
                                 self.optimizer.zero_grad(set_to_none=True)
                             loss.backward()
                             if self.grad_reducer is not None:
                                 self.grad_reducer.reduce()
+                            elif hasattr(self.optimizer, "reduce") and self.world > 1:
+                                self.optimizer.reduce()      # one flat RCCL all-reduce
                             self.optimizer.step()
                     running_loss[mode] += float(loss.detach()) * y_true.shape[0]
                     step += y_true.shape[0]
