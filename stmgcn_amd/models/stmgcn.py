@@ -28,6 +28,20 @@ from ..graph.preprocess import CSRSupport
 
 AdjLike = Union[torch.Tensor, CSRSupport]
 
+# Branch-concurrency stream pool: the M graph branches are independent
+# (reference runs them sequentially, STMGCN.py:112-115); on GPU each branch
+# runs on its own HIP stream so the small gconv/gate kernels of one branch
+# overlap the big persistent-RNN kernels of another. torch autograd replays
+# each op's backward on its forward stream, so the overlap carries to the
+# backward pass; fork/join events keep this capturable in a hipGraph.
+_STREAMS: List[torch.cuda.Stream] = []
+
+
+def _branch_streams(n: int) -> List[torch.cuda.Stream]:
+    while len(_STREAMS) < n:
+        _STREAMS.append(torch.cuda.Stream())
+    return _STREAMS[:n]
+
 
 class GCN(nn.Module):
     """K-support graph convolution op (reference GCN.py:7-46).
@@ -186,9 +200,25 @@ class ST_MGCN(nn.Module):
         if len(sta_adj_list) != self.M:
             raise ValueError(f"expected {self.M} adjacencies, got {len(sta_adj_list)}")
         feat_list = []
-        for m in range(self.M):
-            h = self.rnn_list[m](sta_adj_list[m], obs_seq)     # (B,N,H)
-            feat_list.append(self.gcn_list[m](sta_adj_list[m], h))  # (B,N,G)
+        if obs_seq.is_cuda and self.M > 1:
+            main = torch.cuda.current_stream()
+            streams = _branch_streams(self.M)
+            fork = torch.cuda.Event()
+            fork.record(main)
+            for m in range(self.M):
+                streams[m].wait_event(fork)
+                with torch.cuda.stream(streams[m]):
+                    obs_seq.record_stream(streams[m])
+                    h = self.rnn_list[m](sta_adj_list[m], obs_seq)
+                    f = self.gcn_list[m](sta_adj_list[m], h)
+                feat_list.append(f)
+            for m in range(self.M):
+                main.wait_stream(streams[m])
+                feat_list[m].record_stream(main)
+        else:
+            for m in range(self.M):
+                h = self.rnn_list[m](sta_adj_list[m], obs_seq)     # (B,N,H)
+                feat_list.append(self.gcn_list[m](sta_adj_list[m], h))  # (B,N,G)
         return ops.branch_fuse_head(feat_list, self.fc.weight, self.fc.bias)  # K7
 
 
