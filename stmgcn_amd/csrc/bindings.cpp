@@ -192,7 +192,7 @@ std::vector<at::Tensor> lstm_fwd(at::Tensor x, std::vector<at::Tensor> w_ih,
   void *hp = nullptr, *cp = nullptr, *gp = nullptr;
   if (training) {
     hseq = at::empty({L, Tst, S_pad, kH}, x.options());
-    cseq = at::empty({L, Tst, S_pad * kH}, x.options().dtype(at::kFloat));
+    cseq = at::empty({L, Tst, S_pad * kH}, x.options());
     gates = at::empty({L, Tst, S_pad * 4 * kH}, x.options());
     hp = hseq.data_ptr(); cp = cseq.data_ptr(); gp = gates.data_ptr();
   }

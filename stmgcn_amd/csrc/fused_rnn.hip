@@ -108,7 +108,8 @@ __global__ void __launch_bounds__(256, 1)
 lstm_fwd_kernel(const T* __restrict__ x,    // (S, Tst, C_in)
                 T* __restrict__ out,        // (S, H) or (S, Tst, H)
                 T* __restrict__ hseq_g,     // (L, Tst, S_pad, H) or null
-                float* __restrict__ cseq_g, // (L, Tst, S_pad*H) frag-native
+                T* __restrict__ cseq_g,     // (L, Tst, S_pad*H) frag-native
+                                            // (LSTM c_t / GRU h_{t-1})
                 T* __restrict__ gates_g,    // (L, Tst, S_pad*4H) frag-native
                 RnnPtrs ptrs,
                 int S, int Tst, int L, int ret_seq) {
@@ -301,14 +302,18 @@ lstm_fwd_kernel(const T* __restrict__ x,    // (S, Tst, C_in)
       for (int m = 0; m < 4; ++m)
           *(((frag*)(gp + (m * 64 + lane) * 16)) + 0) = *(frag*)&gsave[m][0],
           *(((frag*)(gp + (m * 64 + lane) * 16)) + 1) = *(frag*)&gsave[m][8];
-        // cell: (L,Tst, S_pad*H) fp32 as [wave][m][lane][4]
-        // (LSTM: c_t; GRU: h_{t-1})
-        float* cp = cseq_g + base * (S_pad * RNN_H)
-                    + (long)blockIdx.x * (SEQ_TILE * RNN_H) + (wv * 4) * 64 * 4;
+        // cell: (L,Tst, S_pad*H) model-dtype as [wave][m][lane][4]
+        // (LSTM: c_t; GRU: h_{t-1}) — T-typed save halves HBM traffic vs
+        // fp32; bwd tolerances cover the rounding (tests at 8% rel)
+        T* cp = cseq_g + base * (S_pad * RNN_H)
+                + (long)blockIdx.x * (SEQ_TILE * RNN_H) + (wv * 4) * 64 * 4;
         #pragma unroll
-      for (int m = 0; m < 4; ++m)
-          *(f32x4*)(cp + (m * 64 + lane) * 4) =
-              f32x4{csave[m][0], csave[m][1], csave[m][2], csave[m][3]};
+      for (int m = 0; m < 4; ++m) {
+          T c4[4];
+          #pragma unroll
+          for (int r = 0; r < 4; ++r) c4[r] = fromF<T>(csave[m][r]);
+          *(ulong1*)(cp + (m * 64 + lane) * 4) = *(ulong1*)c4;
+        }
         // hseq natural layout copy of slot t (also next layer's input source)
         T* hp = hseq_g + base * (S_pad * RNN_H) + (long)(s0)*RNN_H;
         char* slot = hseq + t * 8192;
@@ -349,7 +354,7 @@ void launch_fwd(hipStream_t stream, const void* x, void* out, void* hseq_g,
   const size_t lds_bytes = (size_t)Tst * 8192 + (cin1 ? Tst * 64 * sizeof(T) : (size_t)Tst * 8192);
   auto go = [&](auto kern) {
     hipLaunchKernelGGL(kern, dim3(nblk), dim3(256), lds_bytes, stream,
-                       (const T*)x, (T*)out, (T*)hseq_g, (float*)cseq_g,
+                       (const T*)x, (T*)out, (T*)hseq_g, (T*)cseq_g,
                        (T*)gates_g, ptrs, S, Tst, L, ret_seq);
   };
   if (cin1 && !gru) go(lstm_fwd_kernel<T, true, false>);
@@ -411,7 +416,7 @@ template <typename T, bool CIN1, bool GRU>
 __global__ void __launch_bounds__(256, 1)
 lstm_bwd_kernel(const T* __restrict__ dout,     // (S,H) or (S,Tst,H)
                 const T* __restrict__ x,        // (S,Tst,Cin)
-                const float* __restrict__ cseq_g,
+                const T* __restrict__ cseq_g,   // model dtype (see fwd)
                 const T* __restrict__ gates_g,
                 RnnPtrs w,                      // w_ih/w_hh = TRANSPOSED (C|H, 4H)
                 T* __restrict__ dx,             // (S,Tst,Cin)
@@ -449,9 +454,9 @@ lstm_bwd_kernel(const T* __restrict__ dout,     // (S,H) or (S,Tst,H)
       const long base = (long)layer * Tst + t;
       const T* gp = gates_g + base * (S_pad * 4 * RNN_H)
                     + (long)blockIdx.x * (SEQ_TILE * 4 * RNN_H) + (wv * 4) * 64 * 16;
-      const float* cp_t = cseq_g + base * (S_pad * RNN_H)
-                          + (long)blockIdx.x * (SEQ_TILE * RNN_H) + (wv * 4) * 64 * 4;
-      const float* cp_p = (t > 0) ? cp_t - (long)(S_pad * RNN_H) : nullptr;
+      const T* cp_t = cseq_g + base * (S_pad * RNN_H)
+                      + (long)blockIdx.x * (SEQ_TILE * RNN_H) + (wv * 4) * 64 * 4;
+      const T* cp_p = (t > 0) ? cp_t - (long)(S_pad * RNN_H) : nullptr;
 
       float wih0[4];
       if (l0cin1) {
@@ -464,9 +469,17 @@ lstm_bwd_kernel(const T* __restrict__ dout,     // (S,H) or (S,Tst,H)
       for (int m = 0; m < 4; ++m) {
         frag gf0 = *(((const frag*)(gp + (m * 64 + lane) * 16)) + 0); // i|f
         frag gf1 = *(((const frag*)(gp + (m * 64 + lane) * 16)) + 1); // g|o
-        f32x4 ct = *(const f32x4*)(cp_t + (m * 64 + lane) * 4);
-        f32x4 cpv = cp_p ? *(const f32x4*)(cp_p + (m * 64 + lane) * 4)
-                         : f32x4{0.f, 0.f, 0.f, 0.f};
+        f32x4 ct, cpv;
+        {
+          frag ctv = {}, cpvv = {};
+          *(ulong1*)&ctv = *(const ulong1*)(cp_t + (m * 64 + lane) * 4);
+          if (cp_p) *(ulong1*)&cpvv = *(const ulong1*)(cp_p + (m * 64 + lane) * 4);
+          #pragma unroll
+          for (int r = 0; r < 4; ++r) {
+            ct[r] = elemF(((typename Frag8<T>::elem*)&ctv)[r]);
+            cpv[r] = cp_p ? elemF(((typename Frag8<T>::elem*)&cpvv)[r]) : 0.f;
+          }
+        }
         #pragma unroll
         for (int r = 0; r < 4; ++r) {
           const int row = 16 * m + 4 * lgrp + r;
@@ -610,7 +623,7 @@ void launch_bwd(hipStream_t stream, const void* dout, const void* x,
   const size_t lds_bytes = (size_t)Tst * 8192 + 64 * 512 + 4 * 64 * sizeof(float);
   auto go = [&](auto kern) {
     hipLaunchKernelGGL(kern, dim3(nblk), dim3(256), lds_bytes, stream,
-                       (const T*)dout, (const T*)x, (const float*)cseq_g,
+                       (const T*)dout, (const T*)x, (const T*)cseq_g,
                        (const T*)gates_g, w, (T*)dx, (T*)dA_g, S, Tst, L,
                        ret_seq);
   };
