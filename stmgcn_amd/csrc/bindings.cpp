@@ -112,16 +112,17 @@ at::Tensor cheb_combine(at::Tensor U, at::Tensor rowptr, at::Tensor colidx,
     return Z;
   }
   auto bk1 = at::empty({B, N, C}, U.options());   // b_{j+1}
-  auto bk2 = at::zeros({B, N, C}, U.options());   // b_{j+2}
+  at::Tensor bk2;                                  // b_{j+2}; empty = virtual 0
   // b_K = U_K  (copy)
   step(dt, rowptr, colidx, vals, none, none, ul(K), bk1.data_ptr(), sz, bz,
        B, N, C, 0.f, 0.f, 1.f);
-  Slice sb1{bk1.data_ptr(), sz, bz}, sb2{bk2.data_ptr(), sz, bz};
+  Slice sb1{bk1.data_ptr(), sz, bz}, sb2 = none;
   for (int j = K - 1; j >= 1; --j) {
     auto bnew = at::empty({B, N, C}, U.options());
-    // b_j = 2 G b_{j+1} - b_{j+2} + U_j   (p1 = b_{j+2}, p2 = U_j)
+    // b_j = 2 G b_{j+1} - b_{j+2} + U_j   (p1 = b_{j+2}, p2 = U_j);
+    // b_{K+1} == 0 -> p1 skipped (beta 0) on the first iteration
     step(dt, rowptr, colidx, vals, sb1, sb2, ul(j), bnew.data_ptr(), sz, bz,
-         B, N, C, 2.f, -1.f, 1.f);
+         B, N, C, 2.f, sb2.ptr ? -1.f : 0.f, 1.f);
     bk2 = bk1; bk1 = bnew;
     sb1 = {bk1.data_ptr(), sz, bz}; sb2 = {bk2.data_ptr(), sz, bz};
   }
@@ -239,9 +240,12 @@ std::vector<at::Tensor> lstm_wgrad(at::Tensor dA, at::Tensor hseq, at::Tensor x)
   const int S = x.size(0), cin = x.size(2);
   TORCH_CHECK(cin == 1 || cin == kH);
   auto fopt = dA.options().dtype(at::kFloat);
-  auto dwih = at::zeros({L, 4 * kH, kH}, fopt);
-  auto dwhh = at::zeros({L, 4 * kH, kH}, fopt);
-  auto db = at::zeros({L, 4 * kH}, fopt);
+  // one fill for all three atomic-accumulated outputs
+  const long n_w = (long)L * 4 * kH * kH;
+  auto flat = at::zeros({2 * n_w + (long)L * 4 * kH}, fopt);
+  auto dwih = flat.narrow(0, 0, n_w).view({L, 4 * kH, kH});
+  auto dwhh = flat.narrow(0, n_w, n_w).view({L, 4 * kH, kH});
+  auto db = flat.narrow(0, 2 * n_w, (long)L * 4 * kH).view({L, 4 * kH});
   stmgcn_lstm_wgrad(stream(), dtype_code(dA), dA.data_ptr(), hseq.data_ptr(),
                     x.data_ptr(), dwih.data_ptr<float>(), dwhh.data_ptr<float>(),
                     db.data_ptr<float>(), R, S_pad, S, Tst, L, cin);
@@ -306,7 +310,7 @@ at::Tensor seqsum_permute_bwd(at::Tensor dxs, int64_t C) {
   TORCH_CHECK(dxs.is_cuda() && dxs.dim() == 3);
   dxs = dxs.contiguous();
   const int B = dxs.size(0), N = dxs.size(1), Tst = dxs.size(2);
-  auto dobs = at::zeros({B, Tst, N, C}, dxs.options());
+  auto dobs = at::empty({B, Tst, N, C}, dxs.options());
   stmgcn_seqsum_permute_bwd(stream(), dtype_code(dxs), dxs.data_ptr(),
                             dobs.data_ptr(), B, Tst, N, (int)C);
   return dobs;
