@@ -18,6 +18,9 @@
 // by lstm_bwd_kernel and does ALL layers in one launch (grid.y = L). The
 // h_{t-1} operand for dw_hh is hseq offset by -S_pad rows (zero for t == 0):
 // the host-side torch.cat shift this replaces was 1.9% of step time.
+// grid.z splits the 256 gate rows into 256/GT slices: GT=128 halves the
+// per-wave accumulator state (full-GT compiled to 126 VGPR + 140 AGPR ->
+// 1 wave/SIMD; measured flat ~400 us across chunk counts = latency-bound).
 
 #include "common.h"
 
@@ -92,20 +95,23 @@ __device__ __forceinline__ typename WFrag8<T>::type frag_from(
 //   db[l]   += colsum(dA_l)                           (4H,)
 // dA: (L, R=Tst*S_pad, 4H) natural; hseq: (L, R, H); x: (S, Tst, Cin).
 // Row r of the flat reduction dim maps to (t = r / S_pad, s = r % S_pad).
-template <typename T, bool CIN1>
+template <typename T, bool CIN1, int GT>
 __global__ void __launch_bounds__(256, 1)
 lstm_wgrad_kernel(const T* __restrict__ dA, const T* __restrict__ hseq,
                   const T* __restrict__ x, float* __restrict__ dwih,
                   float* __restrict__ dwhh, float* __restrict__ db,
                   long R, long S_pad, int S, int Tst, int L) {
   using frag = typename WFrag8<T>::type;
+  constexpr int MTG = GT / 64;      // gate m-tiles per wave (GT=128 -> 2)
+  constexpr int GU = (GT / 8) * 16; // dA staging units per K-tile
   extern __shared__ char lds[];
-  char* dAT = lds;                 // [256][32] T -> 16 KiB
-  char* hpT = lds + 16384;         // [64][32]  T ->  4 KiB
-  char* hxT = lds + 16384 + 4096;  // [64][32]  T ->  4 KiB
-  float* red = (float*)lds;        // db reduction scratch (reuses dAT)
+  char* dAT = lds;                     // [GT][32] T
+  char* hpT = lds + GT * 64;           // [64][32] T -> 4 KiB
+  char* hxT = lds + GT * 64 + 4096;    // [64][32] T -> 4 KiB
+  float* red = (float*)lds;            // db reduction scratch (reuses dAT)
 
   const int layer = blockIdx.y;
+  const int g0 = blockIdx.z * GT;      // gate-row slice of this WG
   const int wv = threadIdx.x >> 6, lane = threadIdx.x & 63;
   const int l16 = lane & 15, lgrp = lane >> 4;
   const long chunk = (R + gridDim.x - 1) / gridDim.x;
@@ -117,28 +123,28 @@ lstm_wgrad_kernel(const T* __restrict__ dA, const T* __restrict__ hseq,
   const T* hx_l = (layer > 0) ? hseq + (long)(layer - 1) * R * 64 : nullptr;
   const bool l0 = (layer == 0);
 
-  f32x4 acc_hh[4][4], acc_ih[4][4];
+  f32x4 acc_hh[MTG][4], acc_ih[MTG][4];
   #pragma unroll
-  for (int mt = 0; mt < 4; ++mt)
+  for (int mt = 0; mt < MTG; ++mt)
     #pragma unroll
     for (int nt = 0; nt < 4; ++nt) {
       acc_hh[mt][nt] = f32x4{0.f, 0.f, 0.f, 0.f};
       acc_ih[mt][nt] = f32x4{0.f, 0.f, 0.f, 0.f};
     }
-  float dbp[2][8];
+  float dbp[GU / 256][8];
   #pragma unroll
-  for (int s_ = 0; s_ < 2; ++s_)
+  for (int s_ = 0; s_ < GU / 256; ++s_)
     #pragma unroll
     for (int j = 0; j < 8; ++j) dbp[s_][j] = 0.f;
 
   for (long kt = r0; kt < r1; kt += 32) {
-    // ---- stage dA^T (+ column-sum partials) ------------------------------
-    for (int u = threadIdx.x, slot = 0; u < 512; u += 256, ++slot) {
+    // ---- stage dA^T slice [g0, g0+GT) (+ column-sum partials) ------------
+    for (int u = threadIdx.x, slot = 0; u < GU; u += 256, ++slot) {
       const int pr = u & 15, cb = u >> 4;
       const long r = kt + pr * 2;
       frag v0 = {}, v1 = {};
-      if (r < r1) v0 = *(const frag*)&dA_l[r * 256 + cb * 8];
-      if (r + 1 < r1) v1 = *(const frag*)&dA_l[(r + 1) * 256 + cb * 8];
+      if (r < r1) v0 = *(const frag*)&dA_l[r * 256 + g0 + cb * 8];
+      if (r + 1 < r1) v1 = *(const frag*)&dA_l[(r + 1) * 256 + g0 + cb * 8];
       #pragma unroll
       for (int j = 0; j < 8; ++j) {
         union { T t2[2]; int i; } pk;
@@ -213,10 +219,10 @@ lstm_wgrad_kernel(const T* __restrict__ dA, const T* __restrict__ hseq,
     }
     __syncthreads();
 
-    // ---- MFMA: wave wv owns gate rows [64*wv, 64*wv+64) ------------------
+    // ---- MFMA: wave wv owns gate rows g0 + [wv*16*MTG, ...) --------------
     #pragma unroll
-    for (int mt = 0; mt < 4; ++mt) {
-      const frag a = frag_from<T>(dAT, wv * 64 + mt * 16 + l16, lgrp);
+    for (int mt = 0; mt < MTG; ++mt) {
+      const frag a = frag_from<T>(dAT, wv * 16 * MTG + mt * 16 + l16, lgrp);
       #pragma unroll
       for (int nt = 0; nt < 4; ++nt)
         acc_hh[mt][nt] = wmfma(a, frag_from<T>(hpT, nt * 16 + l16, lgrp),
@@ -237,12 +243,12 @@ lstm_wgrad_kernel(const T* __restrict__ dA, const T* __restrict__ hseq,
   float* whh = dwhh + (long)layer * 256 * 64;
   float* wih = dwih + (long)layer * 256 * 64;
   #pragma unroll
-  for (int mt = 0; mt < 4; ++mt)
+  for (int mt = 0; mt < MTG; ++mt)
     #pragma unroll
     for (int nt = 0; nt < 4; ++nt)
       #pragma unroll
       for (int rr = 0; rr < 4; ++rr) {
-        const int m = wv * 64 + mt * 16 + lgrp * 4 + rr;
+        const int m = g0 + wv * 16 * MTG + mt * 16 + lgrp * 4 + rr;
         const int n = nt * 16 + l16;
         unsafeAtomicAdd(&whh[m * 64 + n], acc_hh[mt][nt][rr]);
         if (!(l0 && CIN1) || nt == 0)
@@ -250,20 +256,22 @@ lstm_wgrad_kernel(const T* __restrict__ dA, const T* __restrict__ hseq,
       }
 
   // ---- db: LDS cross-thread reduction, then atomics ----------------------
+  __syncthreads();      // all MFMA reads of dAT done before red reuses it
   #pragma unroll
-  for (int s_ = 0; s_ < 2; ++s_)
+  for (int s_ = 0; s_ < GU / 256; ++s_)
     #pragma unroll
     for (int j = 0; j < 8; ++j)
       red[(s_ * 256 + threadIdx.x) * 8 + j] = dbp[s_][j];
   __syncthreads();
-  {
-    const int g = threadIdx.x;             // one gate per thread
-    const int gb = g >> 3, j = g & 7;
+  if (threadIdx.x < GT) {
+    const int g = threadIdx.x;             // local gate within the slice
+    const int gb = g >> 3, j = g & 7;      // staging col-block of gate g
+    // contributors: threads whose unit had cb == gb (16 pr values)
     const int set = gb >> 4, base = (gb & 15) * 16;
     float v = 0.f;
     #pragma unroll
     for (int k = 0; k < 16; ++k) v += red[(set * 256 + base + k) * 8 + j];
-    unsafeAtomicAdd(&db[layer * 256 + g], v);
+    unsafeAtomicAdd(&db[layer * 256 + g0 + g], v);
   }
 }
 
@@ -276,31 +284,32 @@ extern "C" void stmgcn_lstm_wgrad(void* stream_v, int dtype, const void* dA,
     const char* e = getenv("STMGCN_WGRAD_CHUNKS");
     env_chunks = e ? atol(e) : 0;
   }
+  constexpr int GT = 128;           // gate-slice width (grid.z = 256/GT)
   const long target = env_chunks > 0 ? env_chunks : 512 / (L > 0 ? L : 1);
   long nchunks = (R + 1023) / 1024;
   if (nchunks > target) nchunks = target;
   if (nchunks < 1) nchunks = 1;
-  const dim3 grid((unsigned)nchunks, L), blk(256);
-  const size_t lds = 16384 + 4096 + 4096;
+  const dim3 grid((unsigned)nchunks, L, 256 / GT), blk(256);
+  const size_t lds = GT * 64 + 4096 + 4096;
   hipStream_t stream = (hipStream_t)stream_v;
   if (dtype == STM_BF16) {
     if (cin == 1)
-      hipLaunchKernelGGL((lstm_wgrad_kernel<__hip_bfloat16, true>), grid, blk,
+      hipLaunchKernelGGL((lstm_wgrad_kernel<__hip_bfloat16, true, GT>), grid, blk,
                          lds, stream, (const __hip_bfloat16*)dA,
                          (const __hip_bfloat16*)hseq, (const __hip_bfloat16*)x,
                          dwih, dwhh, db, R, S_pad, S, Tst, L);
     else
-      hipLaunchKernelGGL((lstm_wgrad_kernel<__hip_bfloat16, false>), grid, blk,
+      hipLaunchKernelGGL((lstm_wgrad_kernel<__hip_bfloat16, false, GT>), grid, blk,
                          lds, stream, (const __hip_bfloat16*)dA,
                          (const __hip_bfloat16*)hseq, (const __hip_bfloat16*)x,
                          dwih, dwhh, db, R, S_pad, S, Tst, L);
   } else if (dtype == STM_F16) {
     if (cin == 1)
-      hipLaunchKernelGGL((lstm_wgrad_kernel<__half, true>), grid, blk, lds,
+      hipLaunchKernelGGL((lstm_wgrad_kernel<__half, true, GT>), grid, blk, lds,
                          stream, (const __half*)dA, (const __half*)hseq,
                          (const __half*)x, dwih, dwhh, db, R, S_pad, S, Tst, L);
     else
-      hipLaunchKernelGGL((lstm_wgrad_kernel<__half, false>), grid, blk, lds,
+      hipLaunchKernelGGL((lstm_wgrad_kernel<__half, false, GT>), grid, blk, lds,
                          stream, (const __half*)dA, (const __half*)hseq,
                          (const __half*)x, dwih, dwhh, db, R, S_pad, S, Tst, L);
   }
