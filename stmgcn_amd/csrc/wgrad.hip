@@ -137,87 +137,82 @@ lstm_wgrad_kernel(const T* __restrict__ dA, const T* __restrict__ hseq,
     #pragma unroll
     for (int j = 0; j < 8; ++j) dbp[s_][j] = 0.f;
 
-  for (long kt = r0; kt < r1; kt += 32) {
-    // ---- stage dA^T slice [g0, g0+GT) (+ column-sum partials) ------------
-    for (int u = threadIdx.x, slot = 0; u < GU; u += 256, ++slot) {
-      const int pr = u & 15, cb = u >> 4;
-      const long r = kt + pr * 2;
-      frag v0 = {}, v1 = {};
-      if (r < r1) v0 = *(const frag*)&dA_l[r * 256 + g0 + cb * 8];
-      if (r + 1 < r1) v1 = *(const frag*)&dA_l[(r + 1) * 256 + g0 + cb * 8];
-      #pragma unroll
-      for (int j = 0; j < 8; ++j) {
-        union { T t2[2]; int i; } pk;
-        pk.t2[0] = ((const T*)&v0)[j];
-        pk.t2[1] = ((const T*)&v1)[j];
-        *(int*)&dAT[tswz(cb * 8 + j, pr * 4)] = pk.i;
-        dbp[slot][j] += toF<T>(pk.t2[0]) + toF<T>(pk.t2[1]);
-      }
-    }
-    // ---- stage h_{t-1}^T: hseq[l] offset by -S_pad rows, zero for t==0 ---
-    // (replaces the host-side torch.cat shift; the t==0 guard must gate the
-    // LOAD — for layer 0, r - S_pad < 0 points before the hseq allocation)
-    for (int u = threadIdx.x; u < 128; u += 256) {
-      const int pr = u & 15, cb = u >> 4;
-      const long ra = kt + pr * 2, rb = ra + 1;
-      frag v0 = {}, v1 = {};
+  // Software-pipelined main loop: global loads for tile k+1 issue right
+  // after the barrier, overlapping tile k's MFMA phase (at 2-3 WGs/CU the
+  // measured kernel was load-latency-bound: flat across chunk counts).
+  static_assert(GU == 256, "one dA unit per thread assumed by the pipeline");
+  const int pr = threadIdx.x & 15, cb = threadIdx.x >> 4;   // dA unit
+  const int hcb = cb & 7;                                   // hp/hx unit (tid<128)
+  const bool hthread = threadIdx.x < 128;
+
+  frag va0, va1, vh0, vh1, vx0, vx1;
+  auto load_tiles = [&](long kt, frag& a0, frag& a1, frag& h0, frag& h1,
+                        frag& x0f, frag& x1f) {
+    const frag fz = {};
+    const long ra = kt + pr * 2, rb = ra + 1;
+    a0 = fz; a1 = fz;
+    if (ra < r1) a0 = *(const frag*)&dA_l[ra * 256 + g0 + cb * 8];
+    if (rb < r1) a1 = *(const frag*)&dA_l[rb * 256 + g0 + cb * 8];
+    h0 = fz; h1 = fz; x0f = fz; x1f = fz;
+    if (hthread) {
+      // h_{t-1}: hseq[l] offset -S_pad rows, zero for t==0 (the guard must
+      // gate the LOAD — for layer 0, r - S_pad points before the allocation)
       if (ra < r1 && ra >= S_pad)
-        v0 = *(const frag*)&hp_l[(ra - S_pad) * 64 + cb * 8];
+        h0 = *(const frag*)&hp_l[(ra - S_pad) * 64 + hcb * 8];
       if (rb < r1 && rb >= S_pad)
-        v1 = *(const frag*)&hp_l[(rb - S_pad) * 64 + cb * 8];
-      #pragma unroll
-      for (int j = 0; j < 8; ++j) {
-        union { T t2[2]; int i; } pk;
-        pk.t2[0] = ((const T*)&v0)[j];
-        pk.t2[1] = ((const T*)&v1)[j];
-        *(int*)&hpT[tswz(cb * 8 + j, pr * 4)] = pk.i;
-      }
-    }
-    // ---- stage layer input^T ---------------------------------------------
-    if (!l0) {
-      stage_tileT<T, 128>(hxT, hx_l, 64, kt, r1, 64, nullptr);
-    } else if (CIN1) {
-      // x (S, Tst, 1): only column 0 live; (t, s) gather
-      for (int u = threadIdx.x; u < 128; u += 256) {
-        const int pr = u & 15, cb = u >> 4;
-        #pragma unroll
-        for (int j = 0; j < 8; ++j) {
-          union { T t2[2]; int i; } pk;
-          pk.t2[0] = fromF<T>(0.f); pk.t2[1] = fromF<T>(0.f);
-          if (cb == 0 && j == 0) {
-            const long ra = kt + pr * 2, rb = ra + 1;
-            long sa = ra % S_pad, ta = ra / S_pad;
-            long sb = rb % S_pad, tb = rb / S_pad;
-            if (ra < r1 && sa < S) pk.t2[0] = x[sa * Tst + ta];
-            if (rb < r1 && sb < S) pk.t2[1] = x[sb * Tst + tb];
-          }
-          *(int*)&hxT[tswz(cb * 8 + j, pr * 4)] = pk.i;
+        h1 = *(const frag*)&hp_l[(rb - S_pad) * 64 + hcb * 8];
+      if (!l0) {
+        if (ra < r1) x0f = *(const frag*)&hx_l[ra * 64 + hcb * 8];
+        if (rb < r1) x1f = *(const frag*)&hx_l[rb * 64 + hcb * 8];
+      } else if (CIN1) {
+        if (hcb == 0) {
+          const long sa = ra % S_pad, ta = ra / S_pad;
+          const long sb = rb % S_pad, tb = rb / S_pad;
+          if (ra < r1 && sa < S) ((T*)&x0f)[0] = x[sa * Tst + ta];
+          if (rb < r1 && sb < S) ((T*)&x1f)[0] = x[sb * Tst + tb];
         }
-      }
-    } else {
-      // x (S, Tst, 64): (t, s) gather with s < S guard
-      for (int u = threadIdx.x; u < 128; u += 256) {
-        const int pr = u & 15, cb = u >> 4;
-        const long ra = kt + pr * 2, rb = ra + 1;
-        frag v0 = {}, v1 = {};
+      } else {
         if (ra < r1) {
           const long s_ = ra % S_pad, t_ = ra / S_pad;
-          if (s_ < S) v0 = *(const frag*)&x[(s_ * Tst + t_) * 64 + cb * 8];
+          if (s_ < S) x0f = *(const frag*)&x[(s_ * Tst + t_) * 64 + hcb * 8];
         }
         if (rb < r1) {
           const long s_ = rb % S_pad, t_ = rb / S_pad;
-          if (s_ < S) v1 = *(const frag*)&x[(s_ * Tst + t_) * 64 + cb * 8];
-        }
-        #pragma unroll
-        for (int j = 0; j < 8; ++j) {
-          union { T t2[2]; int i; } pk;
-          pk.t2[0] = ((const T*)&v0)[j];
-          pk.t2[1] = ((const T*)&v1)[j];
-          *(int*)&hxT[tswz(cb * 8 + j, pr * 4)] = pk.i;
+          if (s_ < S) x1f = *(const frag*)&x[(s_ * Tst + t_) * 64 + hcb * 8];
         }
       }
     }
+  };
+  auto commit_tiles = [&]() {
+    #pragma unroll
+    for (int j = 0; j < 8; ++j) {
+      union { T t2[2]; int i; } pk;
+      pk.t2[0] = ((const T*)&va0)[j];
+      pk.t2[1] = ((const T*)&va1)[j];
+      *(int*)&dAT[tswz(cb * 8 + j, pr * 4)] = pk.i;
+      dbp[0][j] += toF<T>(pk.t2[0]) + toF<T>(pk.t2[1]);
+    }
+    if (hthread) {
+      #pragma unroll
+      for (int j = 0; j < 8; ++j) {
+        union { T t2[2]; int i; } pk;
+        pk.t2[0] = ((const T*)&vh0)[j];
+        pk.t2[1] = ((const T*)&vh1)[j];
+        *(int*)&hpT[tswz(hcb * 8 + j, pr * 4)] = pk.i;
+        pk.t2[0] = ((const T*)&vx0)[j];
+        pk.t2[1] = ((const T*)&vx1)[j];
+        *(int*)&hxT[tswz(hcb * 8 + j, pr * 4)] = pk.i;
+      }
+    }
+  };
+
+  load_tiles(r0, va0, va1, vh0, vh1, vx0, vx1);
+  for (long kt = r0; kt < r1; kt += 32) {
+    commit_tiles();
     __syncthreads();
+    frag na0, na1, nh0, nh1, nx0, nx1;
+    const bool more = kt + 32 < r1;
+    if (more) load_tiles(kt + 32, na0, na1, nh0, nh1, nx0, nx1);
 
     // ---- MFMA: wave wv owns gate rows g0 + [wv*16*MTG, ...) --------------
     #pragma unroll
@@ -236,7 +231,10 @@ lstm_wgrad_kernel(const T* __restrict__ dA, const T* __restrict__ hseq,
                                  acc_ih[mt][nt]);
       }
     }
-    __syncthreads();  // WAR: next stage overwrites the tiles
+    __syncthreads();  // WAR: next commit overwrites the tiles
+    if (more) {
+      va0 = na0; va1 = na1; vh0 = nh0; vh1 = nh1; vx0 = nx0; vx1 = nx1;
+    }
   }
 
   // ---- writeback: one f32 atomic per output element per workgroup --------
