@@ -456,16 +456,30 @@ lstm_bwd_kernel(const T* __restrict__ dout,     // (S,H) or (S,Tst,H)
       #pragma unroll
       for (int r = 0; r < 4; ++r) { dh_rec[m][r] = 0.f; dc[m][r] = 0.f; }
 
+    // ---- training-save prefetch pipeline: gates/cseq for step t are
+    // loaded during step t+1's GEMM phase (the kernel is latency-bound at
+    // 2 WGs/CU; these are the only global loads in the hot loop) ----------
+    const long lay_base = (long)layer * Tst;
+    auto g_at = [&](int t) {
+      return gates_g + (lay_base + t) * (S_pad * 4 * RNN_H)
+             + (long)blockIdx.x * (ST * 4 * RNN_H) + (wv * MT) * 64 * 16;
+    };
+    auto c_at = [&](int t) {
+      return cseq_g + (lay_base + t) * (S_pad * RNN_H)
+             + (long)blockIdx.x * (ST * RNN_H) + (wv * MT) * 64 * 4;
+    };
+    frag gcur[MT][2];            // gates for step t: [m][{i|f, g|o}]
+    #pragma unroll
+    for (int m = 0; m < MT; ++m) {
+      gcur[m][0] = *(((const frag*)(g_at(Tst - 1) + (m * 64 + lane) * 16)) + 0);
+      gcur[m][1] = *(((const frag*)(g_at(Tst - 1) + (m * 64 + lane) * 16)) + 1);
+    }
+
     for (int t = Tst - 1; t >= 0; --t) {
       // WAR: previous step's reads of dA_lds must be complete
       __syncthreads();
 
-      const long base = (long)layer * Tst + t;
-      const T* gp = gates_g + base * (S_pad * 4 * RNN_H)
-                    + (long)blockIdx.x * (ST * 4 * RNN_H) + (wv * MT) * 64 * 16;
-      const T* cp_t = cseq_g + base * (S_pad * RNN_H)
-                      + (long)blockIdx.x * (ST * RNN_H) + (wv * MT) * 64 * 4;
-      const T* cp_p = (t > 0) ? cp_t - (long)(S_pad * RNN_H) : nullptr;
+      const long base = lay_base + t;
 
       float wih0[4];
       if (l0cin1) {
@@ -476,17 +490,18 @@ lstm_bwd_kernel(const T* __restrict__ dout,     // (S,H) or (S,Tst,H)
 
       #pragma unroll
       for (int m = 0; m < MT; ++m) {
-        frag gf0 = *(((const frag*)(gp + (m * 64 + lane) * 16)) + 0); // i|f
-        frag gf1 = *(((const frag*)(gp + (m * 64 + lane) * 16)) + 1); // g|o
+        frag gf0 = gcur[m][0];                                       // i|f
+        frag gf1 = gcur[m][1];                                       // g|o
         f32x4 ct, cpv;
         {
           frag ctv = {}, cpvv = {};
-          *(ulong1*)&ctv = *(const ulong1*)(cp_t + (m * 64 + lane) * 4);
-          if (cp_p) *(ulong1*)&cpvv = *(const ulong1*)(cp_p + (m * 64 + lane) * 4);
+          *(ulong1*)&ctv = *(const ulong1*)(c_at(t) + (m * 64 + lane) * 4);
+          if (t > 0)
+            *(ulong1*)&cpvv = *(const ulong1*)(c_at(t - 1) + (m * 64 + lane) * 4);
           #pragma unroll
           for (int r = 0; r < 4; ++r) {
             ct[r] = elemF(((typename Frag8<T>::elem*)&ctv)[r]);
-            cpv[r] = cp_p ? elemF(((typename Frag8<T>::elem*)&cpvv)[r]) : 0.f;
+            cpv[r] = (t > 0) ? elemF(((typename Frag8<T>::elem*)&cpvv)[r]) : 0.f;
           }
         }
         #pragma unroll
@@ -537,6 +552,16 @@ lstm_bwd_kernel(const T* __restrict__ dout,     // (S,H) or (S,Tst,H)
         }
       }
       __syncthreads();   // dA visible to all waves
+
+      // prefetch next step's gates (overlaps the dA stream + GEMMs below)
+      frag gnext[MT][2];
+      if (t > 0) {
+        #pragma unroll
+        for (int m = 0; m < MT; ++m) {
+          gnext[m][0] = *(((const frag*)(g_at(t - 1) + (m * 64 + lane) * 16)) + 0);
+          gnext[m][1] = *(((const frag*)(g_at(t - 1) + (m * 64 + lane) * 16)) + 1);
+        }
+      }
 
       // ---- stream dA to global (natural layout) for the host-side wgrad --
       {
@@ -617,6 +642,14 @@ lstm_bwd_kernel(const T* __restrict__ dout,     // (S,H) or (S,Tst,H)
             const float v = red[sA] + red[ST + sA] + red[2 * ST + sA] + red[3 * ST + sA];
             if (s0 + sA < S) dx[(long)(s0 + sA) * Tst + t] = fromF<T>(v);
           }
+        }
+      }
+
+      if (t > 0) {
+        #pragma unroll
+        for (int m = 0; m < MT; ++m) {
+          gcur[m][0] = gnext[m][0];
+          gcur[m][1] = gnext[m][1];
         }
       }
     }  // t loop
