@@ -45,11 +45,14 @@ struct RnnPtrs {
 };
 
 // ---------------------------------------------------------------------------
-// LDS swizzle: element (row s in [0,64), channel c in [0,64)) of one 8 KiB
-// timestep slot. Byte = s*128 + c*2, XOR'd so the 16 lanes of a ds_read_b128
-// group (consecutive rows, same channel block) hit distinct banks.
+// LDS swizzle: element (row s, channel c) of one timestep slot (128 B/row).
+// Byte = s*128 + c*2, XOR'd so the 16 lanes of a ds_read_b128 group
+// (consecutive rows, same channel window) hit 64 distinct banks: row parity
+// gives the 32-bank half, ((s>>1)&7) permutes the eight 16 B windows within
+// it — rows s and s+8 land in different windows (the previous (s&7) swizzle
+// collided them 2-way).
 __device__ __forceinline__ int lds_swz(int s, int cbyte) {
-  return s * 128 + (cbyte ^ ((s & 7) << 4));
+  return s * 128 + (cbyte ^ ((((unsigned)s >> 1) & 7) << 4));
 }
 
 template <typename T> struct Frag8;
@@ -320,13 +323,14 @@ lstm_fwd_kernel(const T* __restrict__ x,    // (S, Tst, C_in)
           *(ulong1*)(cp + (m * 64 + lane) * 4) = *(ulong1*)c4;
         }
         // hseq natural layout copy of slot t (also next layer's input source)
+        // (no trailing barrier: slot t's next write is the next layer's
+        // stage t, separated by that stage's pre-write barrier)
         T* hp = hseq_g + base * (S_pad * RNN_H) + (long)(s0)*RNN_H;
         char* slot = hseq + t * SLOT;
         for (int i = threadIdx.x; i < ST * 8; i += 256) {
           const int c8 = i & 7, s = i >> 3;
           *(frag*)&hp[s * RNN_H + c8 * 8] = *(frag*)&slot[lds_swz(s, c8 * 16)];
         }
-        __syncthreads();
       }
     }
   }
