@@ -61,6 +61,10 @@ def main():
     env = init_distributed()
     rank, world = env["rank"], env["world_size"]
     use_gpu = torch.cuda.is_available()
+    if args.graph == "auto" and world > 1:
+        # capture of the RCCL all-reduce is untestable on the 1-GPU CI boxes;
+        # default to eager for multi-rank runs (--graph on forces capture)
+        args.graph = "off"
     device = torch.device(f"cuda:{env['local_rank']}") if use_gpu else torch.device("cpu")
     if use_gpu:
         torch.cuda.set_device(device)

@@ -73,11 +73,12 @@ class ModelTrainer:
         if self.world <= 1:
             return value
         import torch.distributed as dist
-        t = torch.tensor([value], dtype=torch.float64)
-        if torch.distributed.is_initialized():
-            dist.all_reduce(t)
-            return float(t.item()) / self.world
-        return value
+        if not dist.is_initialized():
+            return value
+        dev = "cuda" if dist.get_backend() == "nccl" else "cpu"
+        t = torch.tensor([value], dtype=torch.float64, device=dev)
+        dist.all_reduce(t)
+        return float(t.item()) / self.world
 
     def _forward(self, x, sta_adj_list):
         return self.model(obs_seq=x, sta_adj_list=sta_adj_list)

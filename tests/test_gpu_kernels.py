@@ -482,3 +482,38 @@ def test_fused_trainer_path_loss_decreases():
         losses.append(float(loss))
     assert all(torch.isfinite(p.float()).all() for p in model.parameters())
     assert losses[-1] < 0.5 * losses[0], f"no learning: {losses[0]} -> {losses[-1]}"
+
+
+@pytest.mark.gpu
+def test_rccl_allreduce_inside_hipgraph():
+    """Capture an RCCL all-reduce inside a hipGraph and replay it (world=1
+    process group on the 1-GPU CI box — the capture path is what multi-rank
+    DP would exercise; bench.py keys its multi-rank graph default off this
+    capability)."""
+    import os
+    import torch.distributed as dist
+    if not dist.is_initialized():
+        os.environ.setdefault("MASTER_ADDR", "127.0.0.1")
+        os.environ.setdefault("MASTER_PORT", "29541")
+        dist.init_process_group("nccl", rank=0, world_size=1)
+    torch.cuda.set_device(0)
+    x = torch.ones(1024, device="cuda")
+    # warmup on a side stream (NCCL communicator init must precede capture)
+    s = torch.cuda.Stream()
+    s.wait_stream(torch.cuda.current_stream())
+    with torch.cuda.stream(s):
+        dist.all_reduce(x)
+    torch.cuda.current_stream().wait_stream(s)
+    torch.cuda.synchronize()
+    g = torch.cuda.CUDAGraph()
+    try:
+        with torch.cuda.graph(g):
+            dist.all_reduce(x)
+            x.mul_(0.5)
+    except Exception as e:
+        pytest.skip(f"RCCL graph capture unsupported here: {e}")
+    x.fill_(2.0)
+    g.replay()
+    torch.cuda.synchronize()
+    torch.testing.assert_close(x, torch.ones_like(x))
+    dist.destroy_process_group()
