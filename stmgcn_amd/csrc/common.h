@@ -23,7 +23,18 @@ template <> __device__ __forceinline__ __hip_bfloat16 fromF<__hip_bfloat16>(floa
 }
 template <> __device__ __forceinline__ __half fromF<__half>(float v) { return __float2half(v); }
 
-__device__ __forceinline__ float stm_sigmoid(float x) { return 1.0f / (1.0f + __expf(-x)); }
+// Fast device transcendentals: v_exp_f32 + v_rcp_f32 (~1 ulp each) instead
+// of libm tanhf (~30 VALU instr) / IEEE divide — the RNN pointwise phase is
+// VALU-bound on these (up to 40 per thread per (layer,t) stage) and the
+// results are rounded to bf16/f16 anyway.
+__device__ __forceinline__ float stm_sigmoid(float x) {
+  return __builtin_amdgcn_rcpf(1.0f + __expf(-x));
+}
+__device__ __forceinline__ float stm_tanh(float x) {
+  // tanh(x) = 2*sigmoid(2x) - 1; clamp keeps exp finite for large |x|
+  const float t = (x > 15.f) ? 15.f : (x < -15.f ? -15.f : x);
+  return fmaf(2.0f, stm_sigmoid(2.0f * t), -1.0f);
+}
 
 #define STM_CHECK_HIP(expr)                                                     \
   do {                                                                          \

@@ -259,7 +259,7 @@ lstm_fwd_kernel(const T* __restrict__ x,    // (S, Tst, C_in)
           if (GRU) {
             // q-slots: gi = r-pre, gf = z-pre, gg = n_input, go = n_hidden
             const float r_ = stm_sigmoid(gi), z_ = stm_sigmoid(gf);
-            const float n_ = tanhf(gg + r_ * go);
+            const float n_ = stm_tanh(gg + r_ * go);
             const float hprev = c_state[m][r];
             csave[m][r] = hprev;                        // bwd needs h_{t-1}
             const float h_ = (1.f - z_) * n_ + z_ * hprev;
@@ -271,11 +271,11 @@ lstm_fwd_kernel(const T* __restrict__ x,    // (S, Tst, C_in)
             gsave[m][3 * 4 + r] = fromF<T>(go);          // Bn = Un h + bhn
           } else {
             const float i_ = stm_sigmoid(gi), f_ = stm_sigmoid(gf);
-            const float g_ = tanhf(gg), o_ = stm_sigmoid(go);
+            const float g_ = stm_tanh(gg), o_ = stm_sigmoid(go);
             const float c_ = f_ * c_state[m][r] + i_ * g_;
             c_state[m][r] = c_;
             csave[m][r] = c_;
-            const float h_ = o_ * tanhf(c_);
+            const float h_ = o_ * stm_tanh(c_);
             hval[m][r] = fromF<T>(h_);
             gsave[m][0 * 4 + r] = fromF<T>(i_);
             gsave[m][1 * 4 + r] = fromF<T>(f_);
@@ -539,7 +539,7 @@ lstm_bwd_kernel(const T* __restrict__ dout,     // (S,H) or (S,Tst,H)
             const float f_ = elemF(gf0[4 + r]);
             const float g_ = elemF(gf1[r]);
             const float o_ = elemF(gf1[4 + r]);
-            const float tc = tanhf(ct[r]);
+            const float tc = stm_tanh(ct[r]);
             float dcv = dc[m][r] + dh * o_ * (1.f - tc * tc);
             dAo = dh * tc * o_ * (1.f - o_);
             dAi = dcv * g_ * i_ * (1.f - i_);
