@@ -61,10 +61,10 @@ def main():
     env = init_distributed()
     rank, world = env["rank"], env["world_size"]
     use_gpu = torch.cuda.is_available()
-    if args.graph == "auto" and world > 1:
-        # capture of the RCCL all-reduce is untestable on the 1-GPU CI boxes;
-        # default to eager for multi-rank runs (--graph on forces capture)
-        args.graph = "off"
+    # RCCL all-reduce inside hipGraph capture is verified by
+    # tests/test_gpu_kernels.py::test_rccl_allreduce_inside_hipgraph, so the
+    # whole-step capture (fwd+loss+bwd+all-reduce+Adam) stays the default for
+    # multi-rank runs too; capture failure falls back to eager below.
     device = torch.device(f"cuda:{env['local_rank']}") if use_gpu else torch.device("cpu")
     if use_gpu:
         torch.cuda.set_device(device)

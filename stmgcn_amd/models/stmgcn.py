@@ -34,7 +34,15 @@ AdjLike = Union[torch.Tensor, CSRSupport]
 # overlap the big persistent-RNN kernels of another. torch autograd replays
 # each op's backward on its forward stream, so the overlap carries to the
 # backward pass; fork/join events keep this capturable in a hipGraph.
+# STMGCN_DETERMINISTIC=1 disables the multi-stream path (SURVEY §5
+# race-detection: a serialized debug mode for bisecting stream-ordering bugs).
+import os as _os
+
 _STREAMS: List[torch.cuda.Stream] = []
+
+
+def deterministic_mode() -> bool:
+    return _os.environ.get("STMGCN_DETERMINISTIC", "0") == "1"
 
 
 def _branch_streams(n: int) -> List[torch.cuda.Stream]:
@@ -200,7 +208,7 @@ class ST_MGCN(nn.Module):
         if len(sta_adj_list) != self.M:
             raise ValueError(f"expected {self.M} adjacencies, got {len(sta_adj_list)}")
         feat_list = []
-        if obs_seq.is_cuda and self.M > 1:
+        if obs_seq.is_cuda and self.M > 1 and not deterministic_mode():
             main = torch.cuda.current_stream()
             streams = _branch_streams(self.M)
             fork = torch.cuda.Event()

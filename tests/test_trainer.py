@@ -118,3 +118,24 @@ def test_main_cli_sparse_path(tmp_path):
         capture_output=True, text=True, timeout=600, cwd=REPO)
     assert out.returncode == 0, out.stderr[-2000:]
     assert (tmp_path / "ST_MGCN_best_model.pkl").exists()
+
+
+def test_step_timer_cpu():
+    """StepTimer works without a GPU (perf_counter fallback) and summarizes."""
+    from stmgcn_amd.utils.profiling import StepTimer
+    t = StepTimer(capacity=4)
+    for _ in range(6):
+        with t:
+            sum(range(1000))
+    ms = t.ms()
+    assert len(ms) == 4 and all(m >= 0 for m in ms)
+    s = t.summary()
+    assert s["n"] == 4 and s["p50_ms"] >= 0
+
+
+def test_deterministic_mode_env(monkeypatch):
+    from stmgcn_amd.models.stmgcn import deterministic_mode
+    monkeypatch.delenv("STMGCN_DETERMINISTIC", raising=False)
+    assert not deterministic_mode()
+    monkeypatch.setenv("STMGCN_DETERMINISTIC", "1")
+    assert deterministic_mode()
