@@ -68,9 +68,12 @@ def main():
             adjs.append(pre.process_csr(a).to(device) if hip_path
                         else pre.process(a).to(device=device, dtype=dtype))
 
-    end_day = args.days - 7
-    dates = ["0101", f"{1:02d}{end_day:02d}", f"{1:02d}{end_day + 1:02d}",
-             f"{1:02d}{args.days - 1:02d}"]
+    from datetime import date, timedelta
+    d0 = date(2017, 1, 1)
+    fmt = lambda d: f"{d.month:02d}{d.day:02d}"
+    train_end = d0 + timedelta(days=args.days - 8)
+    dates = ["0101", fmt(train_end), fmt(train_end + timedelta(days=1)),
+             fmt(d0 + timedelta(days=args.days - 2))]
     gen = DataGenerator(dt=cfg.dt, obs_len=tuple(cfg.obs_len),
                         train_test_dates=dates, val_ratio=0.2)
     loaders = gen.get_data_loader(data, cfg.batch_size, device, dtype=dtype)

@@ -38,11 +38,15 @@ class ChebGconvFn(torch.autograd.Function):
                          csr.K_supports, csr.kind == "single")
         B_, N, K, Cin = S.shape
         feat = S.view(B_, N, K * Cin)
-        y = feat @ W.to(feat.dtype)
+        Wd = W.to(feat.dtype)
         if b is not None:
-            y = y + b
+            # bias rides the hipBLASLt GEMM epilogue (one launch, no add)
+            y = torch.addmm(b.to(feat.dtype), feat.view(-1, K * Cin), Wd)
+            y = y.view(B_, N, -1)
+        else:
+            y = feat @ Wd
         if activation == "relu":
-            y = torch.relu(y)
+            y = torch.relu_(y)
         ctx.save_for_backward(feat, W, y if activation == "relu" else None)
         ctx.csr = csr
         ctx.act = activation
