@@ -9,9 +9,9 @@ from __future__ import annotations
 import argparse
 import json
 import os
-
 import sys
-sys.path.insert(0, __import__("os").path.dirname(__import__("os").path.dirname(__import__("os").path.abspath(__file__))))
+
+sys.path.insert(0, os.path.dirname(os.path.dirname(os.path.abspath(__file__))))
 
 import torch
 

@@ -1,12 +1,16 @@
 """autograd.Function wrappers over the CDNA4 HIP kernels (stmgcn_amd._C).
 
-Status (round 1, in progress):
-  - ChebGconvFn: HIP cheb_apply / cheb_combine (in-kernel support recurrence,
-    SURVEY K1) + rocBLAS mix GEMMs. DONE.
-  - FusedRNNFn / contextual_gate / branch_fuse_head: fused kernels in
-    fused_rnn.hip / cg_gate.hip — being brought up; the interim GPU path
-    composes torch ops so the end-to-end GPU slice runs (cheb kernels are
-    already the native load-bearing path).
+Kernel coverage (all landed; SURVEY §2.4 op numbers in parentheses):
+  - ChebGconvFn (K1/K2/K10): in-kernel CSR Chebyshev support recurrence +
+    hipBLASLt mix GEMM with bias epilogue; backward = Clenshaw over G^T and
+    the atb_wgrad reduction-GEMM kernel for dW/db.
+  - FusedLSTMFn / FusedRNNFn (K5/K6/K10): persistent multi-layer LSTM/GRU
+    (fused_rnn.hip) + one batched wgrad launch for ALL layer weight grads
+    (wgrad.hip). GRU rides the 4-slot packing documented on FusedLSTMFn.
+  - SeqsumPermuteFn (K3), GateFn (K4), HeadFn (K7), FusedMSELossFn (K8);
+    the K9 Adam kernel is driven by train/fused_adam.py.
+Numerics tests compare every Function against the fp32 oracle in
+reference_impl.py (tests/test_gpu_kernels.py).
 """
 from __future__ import annotations
 
