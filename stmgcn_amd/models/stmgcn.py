@@ -253,14 +253,34 @@ class STMGCNBlock(nn.Module):
             for _ in range(M)])
 
     def forward(self, x: torch.Tensor, sta_adj_list: Sequence[AdjLike]) -> torch.Tensor:
-        """x: (B, T, N, C) -> (B, T, N, G) (sum-fused over branches)."""
+        """x: (B, T, N, C) -> (B, T, N, G) (sum-fused over branches;
+        branch-concurrent HIP streams as in ST_MGCN.forward)."""
         B, T, N, _ = x.shape
-        out = None
-        for m in range(self.M):
+
+        def branch(m):
             seq = self.rnn_list[m](sta_adj_list[m], x, return_sequences=True)  # (B,T,N,H)
             flat = seq.reshape(B * T, N, seq.shape[-1])
-            g = self.gcn_list[m](sta_adj_list[m], flat).reshape(B, T, N, -1)
-            out = g if out is None else out + g
+            return self.gcn_list[m](sta_adj_list[m], flat).reshape(B, T, N, -1)
+
+        outs = []
+        if x.is_cuda and self.M > 1 and not deterministic_mode():
+            main = torch.cuda.current_stream()
+            streams = _branch_streams(self.M)
+            fork = torch.cuda.Event()
+            fork.record(main)
+            for m in range(self.M):
+                streams[m].wait_event(fork)
+                with torch.cuda.stream(streams[m]):
+                    x.record_stream(streams[m])
+                    outs.append(branch(m))
+            for m in range(self.M):
+                main.wait_stream(streams[m])
+                outs[m].record_stream(main)
+        else:
+            outs = [branch(m) for m in range(self.M)]
+        out = outs[0]
+        for g in outs[1:]:
+            out = out + g
         return out
 
 
