@@ -322,7 +322,8 @@ std::vector<at::Tensor> gate_fwd(at::Tensor obs, at::Tensor g, at::Tensor xs,
   const int B = obs.size(0), Tst = obs.size(1), N = obs.size(2), C = obs.size(3);
   TORCH_CHECK(Tst <= 16, "gate kernel serves T <= 16");
   auto fopt = obs.options().dtype(at::kFloat);
-  auto z = at::empty({B, Tst}, fopt), u = at::empty({B, Tst}, fopt),
+  // z doubles as the multi-block zsum atomic target -> zeroed
+  auto z = at::zeros({B, Tst}, fopt), u = at::empty({B, Tst}, fopt),
        s = at::empty({B, Tst}, fopt);
   auto out = at::empty_like(obs);
   stmgcn_gate_fwd(stream(), dtype_code(obs), g.contiguous().data_ptr(),
@@ -338,7 +339,8 @@ std::vector<at::Tensor> gate_bwd(at::Tensor dout, at::Tensor obs, at::Tensor fcw
   dout = dout.contiguous();
   const int B = obs.size(0), Tst = obs.size(1), N = obs.size(2), C = obs.size(3);
   auto fopt = obs.options().dtype(at::kFloat);
-  auto dz = at::empty({B, Tst}, fopt);
+  // dz doubles as the multi-block ds atomic target -> zeroed
+  auto dz = at::zeros({B, Tst}, fopt);
   auto dw_part = at::empty({B, Tst, Tst}, fopt);
   auto db_part = at::empty({B, Tst}, fopt);
   auto dobs = at::empty_like(obs);

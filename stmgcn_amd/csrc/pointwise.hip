@@ -22,15 +22,22 @@ template <typename T>
 __global__ void seqsum_permute_kernel(const T* __restrict__ obs,  // (B,T,N,C)
                                       T* __restrict__ out,        // (B,N,T)
                                       int B, int Tst, int N, int C) {
-  const long i = (long)blockIdx.x * blockDim.x + threadIdx.x;  // over B*N*T
-  if (i >= (long)B * N * Tst) return;
-  const int t = i % Tst;
-  const int n = (i / Tst) % N;
-  const long b = i / ((long)Tst * N);
+  // iterate in (b,t,n) source order: consecutive threads read consecutive
+  // C-runs (fully coalesced); the (B,N,T) write is a small-stride scatter
+  const long i = (long)blockIdx.x * blockDim.x + threadIdx.x;  // over B*T*N
+  if (i >= (long)B * Tst * N) return;
+  const int n = i % N;
+  const int t = (i / N) % Tst;
+  const long b = i / ((long)N * Tst);
   float acc = 0.f;
-  const T* src = obs + ((b * Tst + t) * (long)N + n) * C;
-  for (int c = 0; c < C; ++c) acc += toF<T>(src[c]);
-  out[i] = fromF<T>(acc);
+  const T* src = obs + i * (long)C;
+  int c = 0;
+  for (; c + 4 <= C; c += 4) {
+    acc += toF<T>(src[c]) + toF<T>(src[c + 1]) + toF<T>(src[c + 2]) +
+           toF<T>(src[c + 3]);
+  }
+  for (; c < C; ++c) acc += toF<T>(src[c]);
+  out[(b * N + n) * (long)Tst + t] = fromF<T>(acc);
 }
 
 template <typename T>
@@ -46,27 +53,27 @@ __global__ void seqsum_permute_bwd_kernel(const T* __restrict__ dxs,  // (B,N,T)
   dobs[i] = dxs[(b * N + n) * (long)Tst + t];
 }
 
-// ---- K4 forward part A: per-batch reduce + tied double-FC + sigmoid -------
-// one block per batch element; saves z, u (post-relu), s for backward.
+// ---- K4 forward part A1: multi-block node reduce -> zsum (B,T) fp32 -------
+// grid (nchunks, B): scales to any N (the old one-block-per-batch form was
+// 3.5 ms/call at N=4096 — 16 blocks on 256 CUs).
 template <typename T>
 __global__ void gate_fwd_reduce_kernel(const T* __restrict__ g,     // (B,N,T)
                                        const T* __restrict__ xs,    // (B,N,T)
-                                       const T* __restrict__ fcw,   // (T,T) row-major
-                                       const T* __restrict__ fcb,   // (T,)
-                                       float* __restrict__ z_out,   // (B,T)
-                                       float* __restrict__ u_out,   // (B,T)
-                                       float* __restrict__ s_out,   // (B,T)
+                                       float* __restrict__ zsum,    // (B,T) zeroed
                                        int N, int Tst) {
   __shared__ float red[256];
-  const long b = blockIdx.x;
+  const long b = blockIdx.y;
+  const int nchunks = gridDim.x;
+  const int chunk = (N + nchunks - 1) / nchunks;
+  const int n0 = blockIdx.x * chunk;
+  const int n1 = min(n0 + chunk, N);
   float zacc[GATE_MAX_T];
   for (int t = 0; t < Tst; ++t) zacc[t] = 0.f;
   const T* gb = g + b * (long)N * Tst;
   const T* xb = xs + b * (long)N * Tst;
-  for (int n = threadIdx.x; n < N; n += blockDim.x)
+  for (int n = n0 + threadIdx.x; n < n1; n += blockDim.x)
     for (int t = 0; t < Tst; ++t)
       zacc[t] += toF<T>(gb[(long)n * Tst + t]) + toF<T>(xb[(long)n * Tst + t]);
-  // block reduce each t
   for (int t = 0; t < Tst; ++t) {
     float v = zacc[t];
     #pragma unroll
@@ -76,25 +83,34 @@ __global__ void gate_fwd_reduce_kernel(const T* __restrict__ g,     // (B,N,T)
     if (threadIdx.x == 0) {
       float tot = 0.f;
       for (int w = 0; w < blockDim.x / 64; ++w) tot += red[w * GATE_MAX_T + t];
-      red[t] = tot / (float)N;                       // z[t]
+      unsafeAtomicAdd(&zsum[b * Tst + t], tot);
     }
     __syncthreads();
   }
-  if (threadIdx.x == 0) {
-    float z[GATE_MAX_T], u[GATE_MAX_T];
-    for (int t = 0; t < Tst; ++t) z[t] = red[t];
-    for (int t = 0; t < Tst; ++t) {                  // u = relu(W z + b)
-      float a = toF<T>(fcb[t]);
-      for (int j = 0; j < Tst; ++j) a += toF<T>(fcw[t * Tst + j]) * z[j];
-      u[t] = a > 0.f ? a : 0.f;
-    }
-    for (int t = 0; t < Tst; ++t) {                  // s = sigmoid(W u + b)
-      float a = toF<T>(fcb[t]);
-      for (int j = 0; j < Tst; ++j) a += toF<T>(fcw[t * Tst + j]) * u[j];
-      z_out[b * Tst + t] = z[t];
-      u_out[b * Tst + t] = u[t];
-      s_out[b * Tst + t] = stm_sigmoid(a);
-    }
+}
+
+// ---- K4 forward part A2: tiny per-batch tied double-FC --------------------
+template <typename T>
+__global__ void gate_fwd_fc_kernel(const float* __restrict__ zsum, // (B,T)
+                                   const T* __restrict__ fcw,      // (T,T)
+                                   const T* __restrict__ fcb,      // (T,)
+                                   float* __restrict__ z_out, float* __restrict__ u_out,
+                                   float* __restrict__ s_out, int N, int Tst, int B) {
+  const long b = (long)blockIdx.x * blockDim.x + threadIdx.x;
+  if (b >= B) return;
+  float z[GATE_MAX_T], u[GATE_MAX_T];
+  for (int t = 0; t < Tst; ++t) z[t] = zsum[b * Tst + t] / (float)N;
+  for (int t = 0; t < Tst; ++t) {                  // u = relu(W z + b)
+    float a = toF<T>(fcb[t]);
+    for (int j = 0; j < Tst; ++j) a += toF<T>(fcw[t * Tst + j]) * z[j];
+    u[t] = a > 0.f ? a : 0.f;
+  }
+  for (int t = 0; t < Tst; ++t) {                  // s = sigmoid(W u + b)
+    float a = toF<T>(fcb[t]);
+    for (int j = 0; j < Tst; ++j) a += toF<T>(fcw[t * Tst + j]) * u[j];
+    z_out[b * Tst + t] = z[t];
+    u_out[b * Tst + t] = u[t];
+    s_out[b * Tst + t] = stm_sigmoid(a);
   }
 }
 
@@ -109,30 +125,26 @@ __global__ void gate_scale_kernel(const T* __restrict__ obs,   // (B,T,N,C)
   out[i] = fromF<T>(toF<T>(obs[i]) * s[bt]);
 }
 
-// ---- K4 backward part A: ds = sum_{n,c} dout*obs; tied-FC backward --------
-// one block per batch; emits dz (B,T) and per-block dW/db partials.
+// ---- K4 backward part A1: ds = sum_{n,c} dout*obs (multi-block + atomics) -
 template <typename T>
 __global__ void gate_bwd_reduce_kernel(const T* __restrict__ dout, // (B,T,N,C)
                                        const T* __restrict__ obs,  // (B,T,N,C)
-                                       const T* __restrict__ fcw,  // (T,T)
-                                       const float* __restrict__ z,
-                                       const float* __restrict__ u,
-                                       const float* __restrict__ s,
-                                       float* __restrict__ dz_out,  // (B,T)
-                                       float* __restrict__ dw_part, // (B,T,T)
-                                       float* __restrict__ db_part, // (B,T)
+                                       float* __restrict__ ds_out, // (B,T) zeroed
                                        int N, int C, int Tst) {
   __shared__ float red[256];
-  const long b = blockIdx.x;
+  const long b = blockIdx.y;
   const long NC = (long)N * C;
+  const int nchunks = gridDim.x;
+  const long chunk = (NC + nchunks - 1) / nchunks;
+  const long i0 = blockIdx.x * chunk;
+  const long i1 = min(i0 + chunk, NC);
   float dsacc[GATE_MAX_T];
   for (int t = 0; t < Tst; ++t) dsacc[t] = 0.f;
   const T* db_ = dout + b * Tst * NC;
   const T* ob = obs + b * Tst * NC;
-  for (long i = threadIdx.x; i < Tst * NC; i += blockDim.x) {
-    const int t = i / NC;
-    dsacc[t] += toF<T>(db_[i]) * toF<T>(ob[i]);
-  }
+  for (int t = 0; t < Tst; ++t)
+    for (long i = i0 + threadIdx.x; i < i1; i += blockDim.x)
+      dsacc[t] += toF<T>(db_[t * NC + i]) * toF<T>(ob[t * NC + i]);
   for (int t = 0; t < Tst; ++t) {
     float v = dsacc[t];
     #pragma unroll
@@ -142,37 +154,48 @@ __global__ void gate_bwd_reduce_kernel(const T* __restrict__ dout, // (B,T,N,C)
     if (threadIdx.x == 0) {
       float tot = 0.f;
       for (int w = 0; w < blockDim.x / 64; ++w) tot += red[w * GATE_MAX_T + t];
-      red[t] = tot;                                   // ds[t]
+      unsafeAtomicAdd(&ds_out[b * Tst + t], tot);
     }
     __syncthreads();
   }
-  if (threadIdx.x == 0) {
-    // tied double-FC backward: s = sig(W u + b), u = relu(W z + b)
-    float dv2[GATE_MAX_T], du[GATE_MAX_T], dv1[GATE_MAX_T], dzv[GATE_MAX_T];
-    for (int t = 0; t < Tst; ++t) {
-      const float sv = s[b * Tst + t];
-      dv2[t] = red[t] * sv * (1.f - sv);
-    }
-    for (int j = 0; j < Tst; ++j) {                   // du = W^T dv2
-      float a = 0.f;
-      for (int t = 0; t < Tst; ++t) a += toF<T>(fcw[t * Tst + j]) * dv2[t];
-      du[j] = a;
-    }
-    for (int t = 0; t < Tst; ++t) dv1[t] = u[b * Tst + t] > 0.f ? du[t] : 0.f;
-    for (int j = 0; j < Tst; ++j) {                   // dz = W^T dv1
-      float a = 0.f;
-      for (int t = 0; t < Tst; ++t) a += toF<T>(fcw[t * Tst + j]) * dv1[t];
-      dzv[j] = a;
-      dz_out[b * Tst + j] = a;
-    }
-    // dW partial = dv2 (x) u + dv1 (x) z ; db partial = dv2 + dv1
-    for (int t = 0; t < Tst; ++t) {
-      for (int j = 0; j < Tst; ++j)
-        dw_part[(b * Tst + t) * Tst + j] =
-            dv2[t] * u[b * Tst + j] + dv1[t] * z[b * Tst + j];
-      db_part[b * Tst + t] = dv2[t] + dv1[t];
-    }
-    (void)dzv;
+}
+
+// ---- K4 backward part A2: tiny per-batch tied-FC backward -----------------
+template <typename T>
+__global__ void gate_bwd_fc_kernel(const float* __restrict__ ds,  // (B,T)
+                                   const T* __restrict__ fcw,     // (T,T)
+                                   const float* __restrict__ z,
+                                   const float* __restrict__ u,
+                                   const float* __restrict__ s,
+                                   float* __restrict__ dz_out,    // (B,T)
+                                   float* __restrict__ dw_part,   // (B,T,T)
+                                   float* __restrict__ db_part,   // (B,T)
+                                   int Tst, int B) {
+  const long b = (long)blockIdx.x * blockDim.x + threadIdx.x;
+  if (b >= B) return;
+  // tied double-FC backward: s = sig(W u + b), u = relu(W z + b)
+  float dv2[GATE_MAX_T], du[GATE_MAX_T], dv1[GATE_MAX_T];
+  for (int t = 0; t < Tst; ++t) {
+    const float sv = s[b * Tst + t];
+    dv2[t] = ds[b * Tst + t] * sv * (1.f - sv);
+  }
+  for (int j = 0; j < Tst; ++j) {                   // du = W^T dv2
+    float a = 0.f;
+    for (int t = 0; t < Tst; ++t) a += toF<T>(fcw[t * Tst + j]) * dv2[t];
+    du[j] = a;
+  }
+  for (int t = 0; t < Tst; ++t) dv1[t] = u[b * Tst + t] > 0.f ? du[t] : 0.f;
+  for (int j = 0; j < Tst; ++j) {                   // dz = W^T dv1
+    float a = 0.f;
+    for (int t = 0; t < Tst; ++t) a += toF<T>(fcw[t * Tst + j]) * dv1[t];
+    dz_out[b * Tst + j] = a;
+  }
+  // dW partial = dv2 (x) u + dv1 (x) z ; db partial = dv2 + dv1
+  for (int t = 0; t < Tst; ++t) {
+    for (int j = 0; j < Tst; ++j)
+      dw_part[(b * Tst + t) * Tst + j] =
+          dv2[t] * u[b * Tst + j] + dv1[t] * z[b * Tst + j];
+    db_part[b * Tst + t] = dv2[t] + dv1[t];
   }
 }
 
@@ -296,6 +319,51 @@ __global__ void adam_prep_kernel(float* hyper) {
   }
 }
 
+// chunks so the node reduction fills the chip regardless of B
+static int gate_chunks(int B, long work_per_b) {
+  long per_chunk = 2048;
+  long nch = (work_per_b + per_chunk - 1) / per_chunk;
+  long cap = (B > 0) ? (2048 / B > 0 ? 2048 / B : 1) : 1;
+  if (nch > cap) nch = cap;
+  if (nch < 1) nch = 1;
+  return (int)nch;
+}
+
+template <typename T>
+static void launch_gate_fwd(hipStream_t st, const T* g, const T* xs,
+                            const T* fcw, const T* fcb, const T* obs,
+                            float* z, float* u, float* s, T* out,
+                            int B, int Tst, int N, int C) {
+  // z arrives zeroed and doubles as the zsum atomic target (A2 rescales)
+  const dim3 gridA(gate_chunks(B, N), B);
+  hipLaunchKernelGGL(gate_fwd_reduce_kernel<T>, gridA, dim3(256), 0, st,
+                     g, xs, z, N, Tst);
+  hipLaunchKernelGGL(gate_fwd_fc_kernel<T>, dim3((B + 63) / 64), dim3(64), 0,
+                     st, z, fcw, fcb, z, u, s, N, Tst, B);
+  hipLaunchKernelGGL(gate_scale_kernel<T>, dim3(((long)B * Tst * N * C + 255) / 256),
+                     dim3(256), 0, st, obs, s, out, Tst, (long)N * C,
+                     (long)B * Tst * N * C);
+}
+
+template <typename T>
+static void launch_gate_bwd(hipStream_t st, const T* dout, const T* obs,
+                            const T* fcw, const float* z, const float* u,
+                            const float* s, float* dz, float* dw_part,
+                            float* db_part, T* dobs, T* dg,
+                            int B, int Tst, int N, int C) {
+  const long total = (long)B * Tst * N * C;
+  const float invN = 1.f / (float)N;
+  // dz arrives zeroed and doubles as the ds atomic target (fc pass rewrites)
+  const dim3 gridA(gate_chunks(B, (long)N * C), B);
+  hipLaunchKernelGGL(gate_bwd_reduce_kernel<T>, gridA, dim3(256), 0, st,
+                     dout, obs, dz, N, C, Tst);
+  hipLaunchKernelGGL(gate_bwd_fc_kernel<T>, dim3((B + 63) / 64), dim3(64), 0,
+                     st, dz, fcw, z, u, s, dz, dw_part, db_part, Tst, B);
+  hipLaunchKernelGGL(gate_bwd_scatter_kernel<T>, dim3((total + 255) / 256),
+                     dim3(256), 0, st, dout, s, dz, dobs, dg, Tst, N, C, invN,
+                     total);
+}
+
 }  // namespace
 
 #define DISPATCH(fn, ...)                                        \
@@ -329,44 +397,40 @@ void stmgcn_seqsum_permute_bwd(void* stream, int dtype, const void* dxs,
   }
 }
 
+
 void stmgcn_gate_fwd(void* stream, int dtype, const void* g, const void* xs,
                      const void* fcw, const void* fcb, const void* obs,
                      float* z, float* u, float* s, void* out,
                      int B, int Tst, int N, int C) {
+  hipStream_t st = (hipStream_t)stream;
   switch (dtype) {
     case STM_F32:
-      hipLaunchKernelGGL(gate_fwd_reduce_kernel<float>, dim3(B), dim3(256), 0, (hipStream_t)stream, (const float*)g, (const float*)xs, (const float*)fcw, (const float*)fcb, z, u, s, N, Tst);
-      hipLaunchKernelGGL(gate_scale_kernel<float>, dim3(((long)B*Tst*N*C + 255)/256), dim3(256), 0, (hipStream_t)stream, (const float*)obs, s, (float*)out, Tst, (long)N*C, (long)B*Tst*N*C);
+      launch_gate_fwd<float>(st, (const float*)g, (const float*)xs, (const float*)fcw, (const float*)fcb, (const float*)obs, z, u, s, (float*)out, B, Tst, N, C);
       break;
     case STM_BF16:
-      hipLaunchKernelGGL(gate_fwd_reduce_kernel<__hip_bfloat16>, dim3(B), dim3(256), 0, (hipStream_t)stream, (const __hip_bfloat16*)g, (const __hip_bfloat16*)xs, (const __hip_bfloat16*)fcw, (const __hip_bfloat16*)fcb, z, u, s, N, Tst);
-      hipLaunchKernelGGL(gate_scale_kernel<__hip_bfloat16>, dim3(((long)B*Tst*N*C + 255)/256), dim3(256), 0, (hipStream_t)stream, (const __hip_bfloat16*)obs, s, (__hip_bfloat16*)out, Tst, (long)N*C, (long)B*Tst*N*C);
+      launch_gate_fwd<__hip_bfloat16>(st, (const __hip_bfloat16*)g, (const __hip_bfloat16*)xs, (const __hip_bfloat16*)fcw, (const __hip_bfloat16*)fcb, (const __hip_bfloat16*)obs, z, u, s, (__hip_bfloat16*)out, B, Tst, N, C);
       break;
     case STM_F16:
-      hipLaunchKernelGGL(gate_fwd_reduce_kernel<__half>, dim3(B), dim3(256), 0, (hipStream_t)stream, (const __half*)g, (const __half*)xs, (const __half*)fcw, (const __half*)fcb, z, u, s, N, Tst);
-      hipLaunchKernelGGL(gate_scale_kernel<__half>, dim3(((long)B*Tst*N*C + 255)/256), dim3(256), 0, (hipStream_t)stream, (const __half*)obs, s, (__half*)out, Tst, (long)N*C, (long)B*Tst*N*C);
+      launch_gate_fwd<__half>(st, (const __half*)g, (const __half*)xs, (const __half*)fcw, (const __half*)fcb, (const __half*)obs, z, u, s, (__half*)out, B, Tst, N, C);
       break;
   }
 }
+
 
 void stmgcn_gate_bwd(void* stream, int dtype, const void* dout, const void* obs,
                      const void* fcw, const float* z, const float* u,
                      const float* s, float* dz, float* dw_part, float* db_part,
                      void* dobs, void* dg, int B, int Tst, int N, int C) {
-  const long total = (long)B * Tst * N * C;
-  const float invN = 1.f / (float)N;
+  hipStream_t st = (hipStream_t)stream;
   switch (dtype) {
     case STM_F32:
-      hipLaunchKernelGGL(gate_bwd_reduce_kernel<float>, dim3(B), dim3(256), 0, (hipStream_t)stream, (const float*)dout, (const float*)obs, (const float*)fcw, z, u, s, dz, dw_part, db_part, N, C, Tst);
-      hipLaunchKernelGGL(gate_bwd_scatter_kernel<float>, dim3((total + 255)/256), dim3(256), 0, (hipStream_t)stream, (const float*)dout, s, dz, (float*)dobs, (float*)dg, Tst, N, C, invN, total);
+      launch_gate_bwd<float>(st, (const float*)dout, (const float*)obs, (const float*)fcw, z, u, s, dz, dw_part, db_part, (float*)dobs, (float*)dg, B, Tst, N, C);
       break;
     case STM_BF16:
-      hipLaunchKernelGGL(gate_bwd_reduce_kernel<__hip_bfloat16>, dim3(B), dim3(256), 0, (hipStream_t)stream, (const __hip_bfloat16*)dout, (const __hip_bfloat16*)obs, (const __hip_bfloat16*)fcw, z, u, s, dz, dw_part, db_part, N, C, Tst);
-      hipLaunchKernelGGL(gate_bwd_scatter_kernel<__hip_bfloat16>, dim3((total + 255)/256), dim3(256), 0, (hipStream_t)stream, (const __hip_bfloat16*)dout, s, dz, (__hip_bfloat16*)dobs, (__hip_bfloat16*)dg, Tst, N, C, invN, total);
+      launch_gate_bwd<__hip_bfloat16>(st, (const __hip_bfloat16*)dout, (const __hip_bfloat16*)obs, (const __hip_bfloat16*)fcw, z, u, s, dz, dw_part, db_part, (__hip_bfloat16*)dobs, (__hip_bfloat16*)dg, B, Tst, N, C);
       break;
     case STM_F16:
-      hipLaunchKernelGGL(gate_bwd_reduce_kernel<__half>, dim3(B), dim3(256), 0, (hipStream_t)stream, (const __half*)dout, (const __half*)obs, (const __half*)fcw, z, u, s, dz, dw_part, db_part, N, C, Tst);
-      hipLaunchKernelGGL(gate_bwd_scatter_kernel<__half>, dim3((total + 255)/256), dim3(256), 0, (hipStream_t)stream, (const __half*)dout, s, dz, (__half*)dobs, (__half*)dg, Tst, N, C, invN, total);
+      launch_gate_bwd<__half>(st, (const __half*)dout, (const __half*)obs, (const __half*)fcw, z, u, s, dz, dw_part, db_part, (__half*)dobs, (__half*)dg, B, Tst, N, C);
       break;
   }
 }
