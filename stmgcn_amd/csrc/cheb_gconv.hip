@@ -40,9 +40,22 @@ __global__ void spmm_step_kernel(
     float acc = 0.0f;
     if (alpha != 0.0f && xb) {
       const int s = rowptr[row], e = rowptr[row + 1];
-      for (int j = s; j < e; ++j) {
-        acc += vals[j] * toF<T>(xb[(long)colidx[j] * sx + c]);
+      // 4-wide unroll: the col/val loads and the x gathers of a chunk issue
+      // together instead of forming one serial load-latency chain per nnz
+      int j = s;
+      for (; j + 4 <= e; j += 4) {
+        int cc[4];
+        float vv[4];
+        #pragma unroll
+        for (int k = 0; k < 4; ++k) { cc[k] = colidx[j + k]; vv[k] = vals[j + k]; }
+        float xv[4];
+        #pragma unroll
+        for (int k = 0; k < 4; ++k) xv[k] = toF<T>(xb[(long)cc[k] * sx + c]);
+        #pragma unroll
+        for (int k = 0; k < 4; ++k) acc = fmaf(vv[k], xv[k], acc);
       }
+      for (; j < e; ++j)
+        acc += vals[j] * toF<T>(xb[(long)colidx[j] * sx + c]);
       acc *= alpha;
     }
     const long ro = (long)b * bo + (long)row * so + c;
