@@ -139,3 +139,29 @@ def test_deterministic_mode_env(monkeypatch):
     assert not deterministic_mode()
     monkeypatch.setenv("STMGCN_DETERMINISTIC", "1")
     assert deterministic_mode()
+
+
+def test_main_cli_end_to_end(tmp_path):
+    """Full CLI path on CPU: synthetic data, 2 epochs, checkpoint written
+    with the reference layout, test metrics printed (reference Main.py flag
+    surface + wiring order)."""
+    import subprocess
+    import sys
+    repo = os.path.dirname(os.path.dirname(os.path.abspath(__file__)))
+    out_dir = tmp_path / "output"
+    r = subprocess.run(
+        [sys.executable, os.path.join(repo, "Main.py"),
+         "--synthetic", "--preset", "cpu-small", "--epochs", "2",
+         "--nodes", "24", "-cpt", "3", "1", "1",
+         "-date", "0101", "0109", "0110", "0112",
+         "--model-dir", str(out_dir),
+         "--metrics", str(tmp_path / "metrics.jsonl")],
+        capture_output=True, text=True, timeout=420, cwd=repo)
+    assert r.returncode == 0, r.stderr[-2000:]
+    assert "Training starts at" in r.stdout
+    assert "true RMSE" in r.stdout
+    ckpt = out_dir / "ST_MGCN_best_model.pkl"
+    assert ckpt.exists()
+    saved = torch.load(ckpt, weights_only=False)
+    assert set(saved) == {"epoch", "state_dict"}
+    assert (tmp_path / "metrics.jsonl").read_text().strip()
