@@ -41,11 +41,20 @@ __device__ __forceinline__ f32x4 wmfma(f16x8 a, f16x8 b, f32x4 c) {
   return __builtin_amdgcn_mfma_f32_16x16x32_f16(a, b, c, 0, 0, 0);
 }
 
-// Transposed-tile LDS addressing: column g (output dim), row-pair byte offset
-// off in [0,64). 64B per column; XOR displaces the 16B fragment window so the
-// MFMA reads (16 consecutive g, 16B each) hit 64 distinct banks.
+// Transposed-tile LDS addressing: column g (output dim), row byte offset off.
+// 32-row tiles (64 B/col): XOR displaces the 16 B fragment window so the MFMA
+// reads (16 consecutive g, 16 B each) hit 64 distinct banks; 64-row tiles
+// (128 B/col) use row parity for the 32-bank half + ((g>>1)&7) window perm.
 __device__ __forceinline__ int tswz(int g, int off) {
   return g * 64 + (off ^ ((((unsigned)g >> 2) & 3) << 4));
+}
+__device__ __forceinline__ int tswz64(int g, int off) {
+  return g * 128 + (off ^ ((((unsigned)g >> 1) & 7) << 4));
+}
+template <typename T>
+__device__ __forceinline__ typename WFrag8<T>::type frag_from64(
+    const char* lds_tile, int g, int koff_bytes) {
+  return *(const typename WFrag8<T>::type*)&lds_tile[tswz64(g, koff_bytes)];
 }
 
 // Stage one 32-row tile of src (rows-major, ld elems) transposed into LDS
@@ -95,22 +104,29 @@ __device__ __forceinline__ typename WFrag8<T>::type frag_from(
 //   db[l]   += colsum(dA_l)                           (4H,)
 // dA: (L, R=Tst*S_pad, 4H) natural; hseq: (L, R, H); x: (S, Tst, Cin).
 // Row r of the flat reduction dim maps to (t = r / S_pad, s = r % S_pad).
-template <typename T, bool CIN1, int GT>
+// L0CIN1 instantiation covers ONLY layer 0 of the scalar-input (C_in == 1)
+// model: its dw_ih is a single column -> acc_ih collapses to one n-tile,
+// which keeps the unified register count at 2 waves/SIMD (the combined
+// variant compiled to 268 regs -> occupancy 1). layer = blockIdx.y + lbase.
+template <typename T, bool L0CIN1, int GT>
 __global__ void __launch_bounds__(256, 1)
 lstm_wgrad_kernel(const T* __restrict__ dA, const T* __restrict__ hseq,
                   const T* __restrict__ x, float* __restrict__ dwih,
                   float* __restrict__ dwhh, float* __restrict__ db,
-                  long R, long S_pad, int S, int Tst, int L) {
+                  long R, long S_pad, int S, int Tst, int L, int lbase) {
   using frag = typename WFrag8<T>::type;
+  constexpr int KT = 64;            // K-tile rows (PMC: 32-row tiles left the
+                                    // waves 65% parked on load latency —
+                                    // 2 MFMAs per frag per barrier now)
   constexpr int MTG = GT / 64;      // gate m-tiles per wave (GT=128 -> 2)
-  constexpr int GU = (GT / 8) * 16; // dA staging units per K-tile
+  constexpr int GU = (GT / 8) * (KT / 2);  // dA staging units per K-tile
   extern __shared__ char lds[];
-  char* dAT = lds;                     // [GT][32] T
-  char* hpT = lds + GT * 64;           // [64][32] T -> 4 KiB
-  char* hxT = lds + GT * 64 + 4096;    // [64][32] T -> 4 KiB
-  float* red = (float*)lds;            // db reduction scratch (reuses dAT)
+  char* dAT = lds;                        // [GT][KT] T
+  char* hpT = lds + GT * 2 * KT;          // [64][KT] T
+  char* hxT = hpT + 64 * 2 * KT;          // [64][KT] T
+  float* red = (float*)lds;               // db reduction scratch (reuses dAT)
 
-  const int layer = blockIdx.y;
+  const int layer = blockIdx.y + lbase;
   const int g0 = blockIdx.z * GT;      // gate-row slice of this WG
   const int wv = threadIdx.x >> 6, lane = threadIdx.x & 63;
   const int l16 = lane & 15, lgrp = lane >> 4;
@@ -121,16 +137,17 @@ lstm_wgrad_kernel(const T* __restrict__ dA, const T* __restrict__ hseq,
   const T* dA_l = dA + (long)layer * R * 256;
   const T* hp_l = hseq + (long)layer * R * 64;       // index r - S_pad
   const T* hx_l = (layer > 0) ? hseq + (long)(layer - 1) * R * 64 : nullptr;
-  const bool l0 = (layer == 0);
+  const bool l0 = L0CIN1 || (layer == 0);
+  constexpr int IHNT = L0CIN1 ? 1 : 4;   // dw_ih n-tiles
 
-  f32x4 acc_hh[MTG][4], acc_ih[MTG][4];
+  f32x4 acc_hh[MTG][4], acc_ih[MTG][IHNT];
   #pragma unroll
-  for (int mt = 0; mt < MTG; ++mt)
+  for (int mt = 0; mt < MTG; ++mt) {
     #pragma unroll
-    for (int nt = 0; nt < 4; ++nt) {
-      acc_hh[mt][nt] = f32x4{0.f, 0.f, 0.f, 0.f};
-      acc_ih[mt][nt] = f32x4{0.f, 0.f, 0.f, 0.f};
-    }
+    for (int nt = 0; nt < 4; ++nt) acc_hh[mt][nt] = f32x4{0.f, 0.f, 0.f, 0.f};
+    #pragma unroll
+    for (int nt = 0; nt < IHNT; ++nt) acc_ih[mt][nt] = f32x4{0.f, 0.f, 0.f, 0.f};
+  }
   float dbp[GU / 256][8];
   #pragma unroll
   for (int s_ = 0; s_ < GU / 256; ++s_)
@@ -138,102 +155,109 @@ lstm_wgrad_kernel(const T* __restrict__ dA, const T* __restrict__ hseq,
     for (int j = 0; j < 8; ++j) dbp[s_][j] = 0.f;
 
   // Software-pipelined main loop: global loads for tile k+1 issue right
-  // after the barrier, overlapping tile k's MFMA phase (at 2-3 WGs/CU the
-  // measured kernel was load-latency-bound: flat across chunk counts).
-  static_assert(GU == 256, "one dA unit per thread assumed by the pipeline");
-  const int pr = threadIdx.x & 15, cb = threadIdx.x >> 4;   // dA unit
-  const int hcb = cb & 7;                                   // hp/hx unit (tid<128)
-  const bool hthread = threadIdx.x < 128;
+  // after the barrier, overlapping tile k's MFMA phase.
+  static_assert(GU == 512, "two dA units per thread assumed by the pipeline");
+  const int pr = threadIdx.x & 31, cb = threadIdx.x >> 5;   // dA unit 0 (of 2)
+  const int pr2 = pr, cb2 = cb + 8;                         // dA unit 1
+  const int hcb = cb;                                       // hp/hx unit (8 blocks)
+  const bool hthread = true;  // all 256 threads carry one hp+hx unit
 
-  frag va0, va1, vh0, vh1, vx0, vx1;
-  auto load_tiles = [&](long kt, frag& a0, frag& a1, frag& h0, frag& h1,
-                        frag& x0f, frag& x1f) {
+  frag va[2][2], vh[2], vx[2];     // [unit][row-pair half] / [row-pair half]
+  auto load_tiles = [&](long kt, frag a[2][2], frag h[2], frag xf[2]) {
     const frag fz = {};
     const long ra = kt + pr * 2, rb = ra + 1;
-    a0 = fz; a1 = fz;
-    if (ra < r1) a0 = *(const frag*)&dA_l[ra * 256 + g0 + cb * 8];
-    if (rb < r1) a1 = *(const frag*)&dA_l[rb * 256 + g0 + cb * 8];
-    h0 = fz; h1 = fz; x0f = fz; x1f = fz;
-    if (hthread) {
-      // h_{t-1}: hseq[l] offset -S_pad rows, zero for t==0 (the guard must
-      // gate the LOAD — for layer 0, r - S_pad points before the allocation)
-      if (ra < r1 && ra >= S_pad)
-        h0 = *(const frag*)&hp_l[(ra - S_pad) * 64 + hcb * 8];
-      if (rb < r1 && rb >= S_pad)
-        h1 = *(const frag*)&hp_l[(rb - S_pad) * 64 + hcb * 8];
-      if (!l0) {
-        if (ra < r1) x0f = *(const frag*)&hx_l[ra * 64 + hcb * 8];
-        if (rb < r1) x1f = *(const frag*)&hx_l[rb * 64 + hcb * 8];
-      } else if (CIN1) {
-        if (hcb == 0) {
-          const long sa = ra % S_pad, ta = ra / S_pad;
-          const long sb = rb % S_pad, tb = rb / S_pad;
-          if (ra < r1 && sa < S) ((T*)&x0f)[0] = x[sa * Tst + ta];
-          if (rb < r1 && sb < S) ((T*)&x1f)[0] = x[sb * Tst + tb];
-        }
-      } else {
-        if (ra < r1) {
-          const long s_ = ra % S_pad, t_ = ra / S_pad;
-          if (s_ < S) x0f = *(const frag*)&x[(s_ * Tst + t_) * 64 + hcb * 8];
-        }
-        if (rb < r1) {
-          const long s_ = rb % S_pad, t_ = rb / S_pad;
-          if (s_ < S) x1f = *(const frag*)&x[(s_ * Tst + t_) * 64 + hcb * 8];
-        }
+    #pragma unroll
+    for (int u = 0; u < 2; ++u) {
+      const int cbu = (u == 0) ? cb : cb2;
+      a[u][0] = fz; a[u][1] = fz;
+      if (ra < r1) a[u][0] = *(const frag*)&dA_l[ra * 256 + g0 + cbu * 8];
+      if (rb < r1) a[u][1] = *(const frag*)&dA_l[rb * 256 + g0 + cbu * 8];
+    }
+    h[0] = fz; h[1] = fz; xf[0] = fz; xf[1] = fz;
+    // h_{t-1}: hseq[l] offset -S_pad rows, zero for t==0 (the guard must
+    // gate the LOAD — for layer 0, r - S_pad points before the allocation)
+    if (ra < r1 && ra >= S_pad)
+      h[0] = *(const frag*)&hp_l[(ra - S_pad) * 64 + hcb * 8];
+    if (rb < r1 && rb >= S_pad)
+      h[1] = *(const frag*)&hp_l[(rb - S_pad) * 64 + hcb * 8];
+    if (!l0) {
+      if (ra < r1) xf[0] = *(const frag*)&hx_l[ra * 64 + hcb * 8];
+      if (rb < r1) xf[1] = *(const frag*)&hx_l[rb * 64 + hcb * 8];
+    } else if (L0CIN1) {
+      if (hcb == 0) {
+        const long sa = ra % S_pad, ta = ra / S_pad;
+        const long sb = rb % S_pad, tb = rb / S_pad;
+        if (ra < r1 && sa < S) ((T*)&xf[0])[0] = x[sa * Tst + ta];
+        if (rb < r1 && sb < S) ((T*)&xf[1])[0] = x[sb * Tst + tb];
+      }
+    } else {
+      if (ra < r1) {
+        const long s_ = ra % S_pad, t_ = ra / S_pad;
+        if (s_ < S) xf[0] = *(const frag*)&x[(s_ * Tst + t_) * 64 + hcb * 8];
+      }
+      if (rb < r1) {
+        const long s_ = rb % S_pad, t_ = rb / S_pad;
+        if (s_ < S) xf[1] = *(const frag*)&x[(s_ * Tst + t_) * 64 + hcb * 8];
       }
     }
   };
   auto commit_tiles = [&]() {
     #pragma unroll
-    for (int j = 0; j < 8; ++j) {
-      union { T t2[2]; int i; } pk;
-      pk.t2[0] = ((const T*)&va0)[j];
-      pk.t2[1] = ((const T*)&va1)[j];
-      *(int*)&dAT[tswz(cb * 8 + j, pr * 4)] = pk.i;
-      dbp[0][j] += toF<T>(pk.t2[0]) + toF<T>(pk.t2[1]);
-    }
-    if (hthread) {
+    for (int u = 0; u < 2; ++u) {
+      const int cbu = (u == 0) ? cb : cb2;
       #pragma unroll
       for (int j = 0; j < 8; ++j) {
         union { T t2[2]; int i; } pk;
-        pk.t2[0] = ((const T*)&vh0)[j];
-        pk.t2[1] = ((const T*)&vh1)[j];
-        *(int*)&hpT[tswz(hcb * 8 + j, pr * 4)] = pk.i;
-        pk.t2[0] = ((const T*)&vx0)[j];
-        pk.t2[1] = ((const T*)&vx1)[j];
-        *(int*)&hxT[tswz(hcb * 8 + j, pr * 4)] = pk.i;
+        pk.t2[0] = ((const T*)&va[u][0])[j];
+        pk.t2[1] = ((const T*)&va[u][1])[j];
+        *(int*)&dAT[tswz64(cbu * 8 + j, pr * 4)] = pk.i;
+        dbp[u][j] += toF<T>(pk.t2[0]) + toF<T>(pk.t2[1]);
       }
+    }
+    #pragma unroll
+    for (int j = 0; j < 8; ++j) {
+      union { T t2[2]; int i; } pk;
+      pk.t2[0] = ((const T*)&vh[0])[j];
+      pk.t2[1] = ((const T*)&vh[1])[j];
+      *(int*)&hpT[tswz64(hcb * 8 + j, pr * 4)] = pk.i;
+      pk.t2[0] = ((const T*)&vx[0])[j];
+      pk.t2[1] = ((const T*)&vx[1])[j];
+      *(int*)&hxT[tswz64(hcb * 8 + j, pr * 4)] = pk.i;
     }
   };
 
-  load_tiles(r0, va0, va1, vh0, vh1, vx0, vx1);
-  for (long kt = r0; kt < r1; kt += 32) {
+  load_tiles(r0, va, vh, vx);
+  for (long kt = r0; kt < r1; kt += KT) {
     commit_tiles();
     __syncthreads();
-    frag na0, na1, nh0, nh1, nx0, nx1;
-    const bool more = kt + 32 < r1;
-    if (more) load_tiles(kt + 32, na0, na1, nh0, nh1, nx0, nx1);
+    frag na[2][2], nh[2], nx[2];
+    const bool more = kt + KT < r1;
+    if (more) load_tiles(kt + KT, na, nh, nx);
 
-    // ---- MFMA: wave wv owns gate rows g0 + [wv*16*MTG, ...) --------------
+    // ---- MFMA: wave wv owns gate rows g0 + [wv*16*MTG, ...); two K-halves
     #pragma unroll
-    for (int mt = 0; mt < MTG; ++mt) {
-      const frag a = frag_from<T>(dAT, wv * 16 * MTG + mt * 16 + l16, lgrp);
+    for (int kh = 0; kh < 2; ++kh) {
+      const int koff = kh * KT + lgrp * 16;   // byte offset of the K half
       #pragma unroll
-      for (int nt = 0; nt < 4; ++nt)
-        acc_hh[mt][nt] = wmfma(a, frag_from<T>(hpT, nt * 16 + l16, lgrp),
-                               acc_hh[mt][nt]);
-      if (l0 && CIN1) {
-        acc_ih[mt][0] = wmfma(a, frag_from<T>(hxT, l16, lgrp), acc_ih[mt][0]);
-      } else {
+      for (int mt = 0; mt < MTG; ++mt) {
+        const frag a = frag_from64<T>(dAT, wv * 16 * MTG + mt * 16 + l16, koff);
         #pragma unroll
         for (int nt = 0; nt < 4; ++nt)
-          acc_ih[mt][nt] = wmfma(a, frag_from<T>(hxT, nt * 16 + l16, lgrp),
+          acc_hh[mt][nt] = wmfma(a, frag_from64<T>(hpT, nt * 16 + l16, koff),
+                                 acc_hh[mt][nt]);
+        #pragma unroll
+        for (int nt = 0; nt < IHNT; ++nt)
+          acc_ih[mt][nt] = wmfma(a, frag_from64<T>(hxT, nt * 16 + l16, koff),
                                  acc_ih[mt][nt]);
       }
     }
     __syncthreads();  // WAR: next commit overwrites the tiles
     if (more) {
-      va0 = na0; va1 = na1; vh0 = nh0; vh1 = nh1; vx0 = nx0; vx1 = nx1;
+      #pragma unroll
+      for (int u = 0; u < 2; ++u) {
+        va[u][0] = na[u][0]; va[u][1] = na[u][1];
+      }
+      vh[0] = nh[0]; vh[1] = nh[1]; vx[0] = nx[0]; vx[1] = nx[1];
     }
   }
 
@@ -249,26 +273,27 @@ lstm_wgrad_kernel(const T* __restrict__ dA, const T* __restrict__ hseq,
         const int m = g0 + wv * 16 * MTG + mt * 16 + lgrp * 4 + rr;
         const int n = nt * 16 + l16;
         unsafeAtomicAdd(&whh[m * 64 + n], acc_hh[mt][nt][rr]);
-        if (!(l0 && CIN1) || nt == 0)
-          unsafeAtomicAdd(&wih[m * 64 + n], acc_ih[mt][nt][rr]);
+        if (nt < IHNT)
+          unsafeAtomicAdd(&wih[m * 64 + n], acc_ih[mt][nt < IHNT ? nt : 0][rr]);
       }
 
   // ---- db: LDS cross-thread reduction, then atomics ----------------------
+  // unit u of thread tid covered gate block (tid>>5) + 8u, rows pr = tid&31
   __syncthreads();      // all MFMA reads of dAT done before red reuses it
   #pragma unroll
-  for (int s_ = 0; s_ < GU / 256; ++s_)
+  for (int u = 0; u < 2; ++u)
     #pragma unroll
     for (int j = 0; j < 8; ++j)
-      red[(s_ * 256 + threadIdx.x) * 8 + j] = dbp[s_][j];
+      red[(u * 256 + threadIdx.x) * 8 + j] = dbp[u][j];
   __syncthreads();
   if (threadIdx.x < GT) {
     const int g = threadIdx.x;             // local gate within the slice
     const int gb = g >> 3, j = g & 7;      // staging col-block of gate g
-    // contributors: threads whose unit had cb == gb (16 pr values)
-    const int set = gb >> 4, base = (gb & 15) * 16;
+    // contributors: 32 threads tid = (gb & 7) * 32 + pr, unit set gb >> 3
+    const int set = gb >> 3, base = (gb & 7) * 32;
     float v = 0.f;
     #pragma unroll
-    for (int k = 0; k < 16; ++k) v += red[(set * 256 + base + k) * 8 + j];
+    for (int k = 0; k < 32; ++k) v += red[(set * 256 + base + k) * 8 + j];
     unsafeAtomicAdd(&db[layer * 256 + g0 + g], v);
   }
 }
@@ -287,30 +312,31 @@ extern "C" void stmgcn_lstm_wgrad(void* stream_v, int dtype, const void* dA,
   long nchunks = (R + 1023) / 1024;
   if (nchunks > target) nchunks = target;
   if (nchunks < 1) nchunks = 1;
-  const dim3 grid((unsigned)nchunks, L, 256 / GT), blk(256);
-  const size_t lds = GT * 64 + 4096 + 4096;
+  const size_t lds = (size_t)GT * 128 + 8192 + 8192;  // 64-row tiles
   hipStream_t stream = (hipStream_t)stream_v;
-  if (dtype == STM_BF16) {
-    if (cin == 1)
-      hipLaunchKernelGGL((lstm_wgrad_kernel<__hip_bfloat16, true, GT>), grid, blk,
-                         lds, stream, (const __hip_bfloat16*)dA,
-                         (const __hip_bfloat16*)hseq, (const __hip_bfloat16*)x,
-                         dwih, dwhh, db, R, S_pad, S, Tst, L);
-    else
-      hipLaunchKernelGGL((lstm_wgrad_kernel<__hip_bfloat16, false, GT>), grid, blk,
-                         lds, stream, (const __hip_bfloat16*)dA,
-                         (const __hip_bfloat16*)hseq, (const __hip_bfloat16*)x,
-                         dwih, dwhh, db, R, S_pad, S, Tst, L);
-  } else if (dtype == STM_F16) {
-    if (cin == 1)
-      hipLaunchKernelGGL((lstm_wgrad_kernel<__half, true, GT>), grid, blk, lds,
-                         stream, (const __half*)dA, (const __half*)hseq,
-                         (const __half*)x, dwih, dwhh, db, R, S_pad, S, Tst, L);
-    else
-      hipLaunchKernelGGL((lstm_wgrad_kernel<__half, false, GT>), grid, blk, lds,
-                         stream, (const __half*)dA, (const __half*)hseq,
-                         (const __half*)x, dwih, dwhh, db, R, S_pad, S, Tst, L);
-  }
+  const dim3 blk(256);
+  auto launch = [&](auto tptr) {
+    using TT = std::remove_pointer_t<decltype(tptr)>;
+    if (cin == 1) {
+      // layer 0 (single ih column) and layers >= 1 as separate instantiations
+      hipLaunchKernelGGL((lstm_wgrad_kernel<TT, true, GT>),
+                         dim3((unsigned)nchunks, 1, 256 / GT), blk, lds, stream,
+                         (const TT*)dA, (const TT*)hseq, (const TT*)x,
+                         dwih, dwhh, db, R, S_pad, S, Tst, L, 0);
+      if (L > 1)
+        hipLaunchKernelGGL((lstm_wgrad_kernel<TT, false, GT>),
+                           dim3((unsigned)nchunks, L - 1, 256 / GT), blk, lds,
+                           stream, (const TT*)dA, (const TT*)hseq, (const TT*)x,
+                           dwih, dwhh, db, R, S_pad, S, Tst, L, 1);
+    } else {
+      hipLaunchKernelGGL((lstm_wgrad_kernel<TT, false, GT>),
+                         dim3((unsigned)nchunks, L, 256 / GT), blk, lds, stream,
+                         (const TT*)dA, (const TT*)hseq, (const TT*)x,
+                         dwih, dwhh, db, R, S_pad, S, Tst, L, 0);
+    }
+  };
+  if (dtype == STM_BF16) launch((__hip_bfloat16*)nullptr);
+  else if (dtype == STM_F16) launch((__half*)nullptr);
 }
 
 // ===========================================================================
