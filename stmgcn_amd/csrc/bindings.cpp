@@ -295,6 +295,8 @@ void stmgcn_mse_fwd(void*, int, const void*, const void*, float*, void*, long);
 void stmgcn_mse_bwd(void*, int, const void*, const float*, void*, long);
 void stmgcn_adam(void*, int, float*, void*, const void*, float*, float*,
                  float*, long);
+void stmgcn_gather_grads(void*, int, const void**, const long*, const int*,
+                         int, void*);
 }
 
 at::Tensor seqsum_permute(at::Tensor obs) {
@@ -398,6 +400,33 @@ at::Tensor mse_bwd(at::Tensor diff, at::Tensor gscale) {
   return dpred;
 }
 
+// Pack per-param grads (or zeros for undefined) into the flat grad arena.
+// grads: list aligned with the arena layout; offsets/lens in elements.
+void gather_grads(std::vector<at::Tensor> grads, std::vector<int64_t> ofs,
+                  std::vector<int64_t> lens, at::Tensor arena) {
+  TORCH_CHECK(arena.is_cuda() && arena.is_contiguous());
+  const int n = (int)ofs.size();
+  const void* srcs[64];
+  long o[64];
+  int l[64];
+  int dt = dtype_code(arena);
+  for (int base = 0; base < n; base += 64) {
+    const int cnt = std::min(64, n - base);
+    for (int i = 0; i < cnt; ++i) {
+      const auto& g = grads[base + i];
+      if (g.defined() && g.numel() > 0) {
+        TORCH_CHECK(g.is_contiguous() && g.scalar_type() == arena.scalar_type());
+        srcs[i] = g.data_ptr();
+      } else {
+        srcs[i] = nullptr;   // no grad flowed -> zero-fill the segment
+      }
+      o[i] = ofs[base + i];
+      l[i] = (int)lens[base + i];
+    }
+    stmgcn_gather_grads(stream(), dt, srcs, o, l, cnt, arena.data_ptr());
+  }
+}
+
 void adam_step(at::Tensor master, at::Tensor param, at::Tensor grad,
                at::Tensor m, at::Tensor v, at::Tensor hyper) {
   TORCH_CHECK(master.is_cuda() && master.scalar_type() == at::kFloat);
@@ -421,6 +450,7 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("mse_fwd", &mse_fwd, "K8: fused MSE loss forward");
   m.def("mse_bwd", &mse_bwd, "K8: fused MSE grad");
   m.def("adam_step", &adam_step, "K9: multi-tensor Adam over flat arena");
+  m.def("gather_grads", &gather_grads, "K9b: pack per-param grads into arena");
   m.def("cheb_apply", &cheb_apply,
         "Support stack S[b,n,k,c] = (T_k(G) x)[b,n,c] via in-kernel recurrence");
   m.def("cheb_combine", &cheb_combine,

@@ -491,3 +491,47 @@ void stmgcn_adam(void* stream, int dtype, float* master, void* param,
 }
 
 }  // extern "C"
+
+// ---- K9b: multi-segment gradient gather into the flat Adam arena ----------
+// FusedAdam runs autograd with .grad = None (no per-param accumulate-add
+// kernels); the fresh grads are packed into the contiguous arena by ONE
+// launch per <=64 segments. Null src -> zero-fill (param got no grad).
+namespace {
+struct GatherSeg {
+  const void* src[64];
+  long ofs[64];
+  int len[64];
+};
+
+template <typename T>
+__global__ void gather_grads_kernel(GatherSeg a, T* __restrict__ arena,
+                                    int nseg) {
+  const int seg = blockIdx.x;
+  if (seg >= nseg) return;
+  T* dst = arena + a.ofs[seg];
+  const T* s = (const T*)a.src[seg];
+  const int len = a.len[seg];
+  const int step = blockDim.x * gridDim.y;
+  for (int i = blockIdx.y * blockDim.x + threadIdx.x; i < len; i += step)
+    dst[i] = s ? s[i] : fromF<T>(0.f);
+}
+}  // namespace
+
+extern "C" void stmgcn_gather_grads(void* stream, int dtype, const void** srcs,
+                                    const long* ofs, const int* lens, int nseg,
+                                    void* arena) {
+  GatherSeg a;
+  int maxlen = 1;
+  for (int i = 0; i < nseg && i < 64; ++i) {
+    a.src[i] = srcs[i]; a.ofs[i] = ofs[i]; a.len[i] = lens[i];
+    if (lens[i] > maxlen) maxlen = lens[i];
+  }
+  const int chunks = min(8, (maxlen + 2047) / 2048);
+  const dim3 grid(nseg, chunks), blk(256);
+  hipStream_t st = (hipStream_t)stream;
+  switch (dtype) {
+    case STM_F32: hipLaunchKernelGGL(gather_grads_kernel<float>, grid, blk, 0, st, a, (float*)arena, nseg); break;
+    case STM_BF16: hipLaunchKernelGGL(gather_grads_kernel<__hip_bfloat16>, grid, blk, 0, st, a, (__hip_bfloat16*)arena, nseg); break;
+    case STM_F16: hipLaunchKernelGGL(gather_grads_kernel<__half>, grid, blk, 0, st, a, (__half*)arena, nseg); break;
+  }
+}

@@ -42,17 +42,18 @@ class FusedAdam:
 
         n = sum(p.numel() for p in self.params)
         self.arena = torch.empty(n, dtype=dt, device=dev)
-        self.grad_arena = torch.zeros(n, dtype=dt, device=dev)
+        self.grad_arena = torch.empty(n, dtype=dt, device=dev)
         ofs = 0
-        self._views = []
+        self._offsets: List[int] = []
+        self._lens: List[int] = []
         for p in self.params:
             k = p.numel()
             self.arena[ofs:ofs + k].copy_(p.data.reshape(-1))
             p.data = self.arena[ofs:ofs + k].view_as(p)
-            g = self.grad_arena[ofs:ofs + k].view_as(p)
-            p.grad = g
-            self._views.append(g)
+            self._offsets.append(ofs)
+            self._lens.append(k)
             ofs += k
+        self._gathered = False
         self.master = self.arena.float()
         self.m = torch.zeros(n, dtype=torch.float32, device=dev)
         self.v = torch.zeros(n, dtype=torch.float32, device=dev)
@@ -63,22 +64,39 @@ class FusedAdam:
         self.param_groups = [{"params": self.params, "lr": lr,
                               "weight_decay": weight_decay}]
 
-    def zero_grad(self, set_to_none: bool = False):
-        self.grad_arena.zero_()
-        for p, g in zip(self.params, self._views):
-            if p.grad is not g:
-                p.grad = g
+    def zero_grad(self, set_to_none: bool = True):
+        """Grads run set-to-none: autograd STORES each produced grad (no
+        per-param accumulate-add kernels — 56 launches/step at reference
+        scale); step()/reduce() pack them into the flat arena with one
+        gather launch per 64 params."""
+        for p in self.params:
+            p.grad = None
+        self._gathered = False
+
+    def _ensure_gathered(self):
+        if self._gathered:
+            return
+        grads = []
+        for p in self.params:
+            g = p.grad
+            if g is not None and not g.is_contiguous():
+                g = g.contiguous()
+            grads.append(g if g is not None else torch.Tensor())
+        self._C.gather_grads(grads, self._offsets, self._lens, self.grad_arena)
+        self._gathered = True
 
     def reduce(self):
         """DP: one flat all-reduce over RCCL + 1/world rescale."""
         if self.world <= 1:
             return
         import torch.distributed as dist
+        self._ensure_gathered()
         dist.all_reduce(self.grad_arena, group=self.group)
         self.grad_arena.mul_(1.0 / self.world)
 
     @torch.no_grad()
     def step(self, closure=None):
+        self._ensure_gathered()
         self._C.adam_step(self.master, self.arena, self.grad_arena, self.m,
                           self.v, self.hyper)
 
