@@ -319,7 +319,7 @@ def test_fused_adam_matches_torch_adam():
         opt_t.step()
         opt_f.zero_grad()
         for p, g in zip(ps_f, gs):
-            p.grad.copy_(g.to(dev))
+            p.grad = g.to(dev).to(p.dtype)
         opt_f.step()
     for pt, pf in zip(ps_t, ps_f):
         torch.testing.assert_close(pf.data, pt.data, rtol=1e-5, atol=1e-6)
