@@ -373,10 +373,62 @@ atb_wgrad_kernel(const T* __restrict__ A, const T* __restrict__ B,
   #pragma unroll
   for (int j = 0; j < 8; ++j) dbp[j] = 0.f;
 
+  // register-prefetch pipeline (same rationale as lstm_wgrad_kernel): A has
+  // up to 2 units/thread (unit u: pair pr = u & 15, col-block cb = u >> 4),
+  // B one unit for tid < 128
+  const int pr = threadIdx.x & 15;
+  const int cba = threadIdx.x >> 4, cba2 = cba + 16;  // A col-blocks
+  const int cbb = cba & 7;                            // B col-block (tid<128)
+  const frag fz = {};
+  frag va[2][2], vb[2];
+  auto load_tiles = [&](long kt, frag a[2][2], frag b[2]) {
+    const long ra = kt + pr * 2, rb = ra + 1;
+    a[0][0] = fz; a[0][1] = fz; a[1][0] = fz; a[1][1] = fz;
+    if (cba * 8 < M) {
+      if (ra < r1) a[0][0] = *(const frag*)&A[ra * M + cba * 8];
+      if (rb < r1) a[0][1] = *(const frag*)&A[rb * M + cba * 8];
+    }
+    if (cba2 * 8 < M) {
+      if (ra < r1) a[1][0] = *(const frag*)&A[ra * M + cba2 * 8];
+      if (rb < r1) a[1][1] = *(const frag*)&A[rb * M + cba2 * 8];
+    }
+    b[0] = fz; b[1] = fz;
+    if (threadIdx.x < 128 && cbb * 8 < N) {
+      if (ra < r1) b[0] = *(const frag*)&B[ra * N + cbb * 8];
+      if (rb < r1) b[1] = *(const frag*)&B[rb * N + cbb * 8];
+    }
+  };
+  auto commit_tiles = [&]() {
+    #pragma unroll
+    for (int u = 0; u < 2; ++u) {
+      const int cb = (u == 0) ? cba : cba2;
+      #pragma unroll
+      for (int j = 0; j < 8; ++j) {
+        union { T t2[2]; int i; } pk;
+        pk.t2[0] = ((const T*)&va[u][0])[j];
+        pk.t2[1] = ((const T*)&va[u][1])[j];
+        *(int*)&AT[tswz(cb * 8 + j, pr * 4)] = pk.i;
+      }
+    }
+    if (threadIdx.x < 128) {
+      #pragma unroll
+      for (int j = 0; j < 8; ++j) {
+        union { T t2[2]; int i; } pk;
+        pk.t2[0] = ((const T*)&vb[0])[j];
+        pk.t2[1] = ((const T*)&vb[1])[j];
+        *(int*)&BT[tswz(cbb * 8 + j, pr * 4)] = pk.i;
+        if (db) dbp[j] += toF<T>(pk.t2[0]) + toF<T>(pk.t2[1]);
+      }
+    }
+  };
+
+  load_tiles(r0, va, vb);
   for (long kt = r0; kt < r1; kt += 32) {
-    stage_tileT<T, 512>(AT, A, M, kt, r1, M, nullptr);
-    stage_tileT<T, 128>(BT, B, N, kt, r1, N, db ? dbp : nullptr);
+    commit_tiles();
     __syncthreads();
+    frag na[2][2], nb[2];
+    const bool more = kt + 32 < r1;
+    if (more) load_tiles(kt + 32, na, nb);
     for (int i = 0; i < mt_mine; ++i) {
       const int mt = wv + i * 4;
       const frag a = frag_from<T>(AT, mt * 16 + l16, lgrp);
@@ -384,6 +436,11 @@ atb_wgrad_kernel(const T* __restrict__ A, const T* __restrict__ B,
         acc[i][nt] = wmfma(a, frag_from<T>(BT, nt * 16 + l16, lgrp), acc[i][nt]);
     }
     __syncthreads();
+    if (more) {
+      va[0][0] = na[0][0]; va[0][1] = na[0][1];
+      va[1][0] = na[1][0]; va[1][1] = na[1][1];
+      vb[0] = nb[0]; vb[1] = nb[1];
+    }
   }
 
   for (int i = 0; i < mt_mine; ++i) {
