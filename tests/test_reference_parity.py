@@ -148,3 +148,21 @@ def test_data_pipeline_window_parity(tmp_path):
     z = np.linspace(-1, 1, 13, dtype=np.float32)
     np.testing.assert_allclose(our_in.minmax_denormalize(z),
                                ref_in.minmax_denormalize(z), rtol=1e-6)
+
+
+def test_metric_formula_parity(ref_modules):
+    """MSE/RMSE/MAE/MAPE(eps=1.0)/PCC match the reference ModelTrainer
+    statics bit-for-bit (Model_Trainer.py:100-114, quirk 11)."""
+    sys.path.insert(0, REF_DIR)
+    try:
+        import Model_Trainer as ref_mt
+    finally:
+        sys.path.remove(REF_DIR)
+    from stmgcn_amd.train import ModelTrainer as Ours
+    rng = np.random.default_rng(4)
+    y_pred = rng.normal(50, 20, (100, 16, 1)).astype(np.float32)
+    y_true = np.abs(rng.normal(50, 20, (100, 16, 1))).astype(np.float32)
+    for name in ["MSE", "RMSE", "MAE", "MAPE", "PCC"]:
+        r = getattr(ref_mt.ModelTrainer, name)(y_pred, y_true)
+        o = getattr(Ours, name)(y_pred, y_true)
+        np.testing.assert_allclose(o, r, rtol=1e-6, err_msg=name)
