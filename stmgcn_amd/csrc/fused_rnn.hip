@@ -567,17 +567,9 @@ lstm_bwd_kernel(const T* __restrict__ dout,     // (S,H) or (S,Tst,H)
         }
       }
 
-      // ---- stream dA to global (natural layout) for the host-side wgrad --
-      {
-        T* out_dA = dA_g + base * (S_pad * 4 * RNN_H) + (long)s0 * 4 * RNN_H;
-        for (int i = threadIdx.x; i < ST * 32; i += 256) {
-          const int c8 = i & 31, sA = i >> 5;   // 32 x 16B pieces per row
-          *(frag*)&out_dA[(long)sA * 4 * RNN_H + c8 * 8] =
-              *(const frag*)&dA_lds[swzA(sA, c8 * 16)];
-        }
-      }
-
       // ---- GEMM1: dh_prev = dA @ W_hh ------------------------------------
+      // (the dA global stream moved below the GEMMs: dh_prev is the serial
+      // BPTT critical path, the stream stores fill the gaps after it)
       if (t > 0) {
         f32x4 acc[MT];
         #pragma unroll
@@ -646,6 +638,16 @@ lstm_bwd_kernel(const T* __restrict__ dout,     // (S,H) or (S,Tst,H)
             const float v = red[sA] + red[ST + sA] + red[2 * ST + sA] + red[3 * ST + sA];
             if (s0 + sA < S) dx[(long)(s0 + sA) * Tst + t] = fromF<T>(v);
           }
+        }
+      }
+
+      // ---- stream dA to global (natural layout) for the wgrad kernel -----
+      {
+        T* out_dA = dA_g + base * (S_pad * 4 * RNN_H) + (long)s0 * 4 * RNN_H;
+        for (int i = threadIdx.x; i < ST * 32; i += 256) {
+          const int c8 = i & 31, sA = i >> 5;   // 32 x 16B pieces per row
+          *(frag*)&out_dA[(long)sA * 4 * RNN_H + c8 * 8] =
+              *(const frag*)&dA_lds[swzA(sA, c8 * 16)];
         }
       }
 
