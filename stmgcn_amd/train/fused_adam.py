@@ -54,6 +54,11 @@ class FusedAdam:
             self._lens.append(k)
             ofs += k
         self._gathered = False
+        if self.world > 1:
+            # identical initial weights on every rank: ONE flat broadcast
+            # (GradReducer does the per-tensor equivalent on the torch path)
+            import torch.distributed as dist
+            dist.broadcast(self.arena, src=0, group=self.group)
         self.master = self.arena.float()
         self.m = torch.zeros(n, dtype=torch.float32, device=dev)
         self.v = torch.zeros(n, dtype=torch.float32, device=dev)
