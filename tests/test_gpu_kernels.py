@@ -517,3 +517,39 @@ def test_rccl_allreduce_inside_hipgraph():
     torch.cuda.synchronize()
     torch.testing.assert_close(x, torch.ones_like(x))
     dist.destroy_process_group()
+
+
+@pytest.mark.gpu
+@pytest.mark.parametrize("cell", ["lstm", "gru"])
+def test_fused_rnn_fp16_forward_backward(cell):
+    """fp16 variants of the fused RNN kernels (the deep-4096 preset dtype)
+    vs the fp32 oracle."""
+    from stmgcn_amd.ops.hip_ops import FusedLSTMFn
+    S, T, L, cin = 96, 8, 2, 64
+    if cell == "lstm":
+        ws = _lstm_oracle_weights(L, cin, seed=9)
+        x = torch.randn(S, T, cin)
+        xr = x.clone().requires_grad_(True)
+        wr = [w.clone().requires_grad_(True) for w in ws]
+        out_ref = ref.lstm_forward(xr, wr, torch.zeros(L, S, 64),
+                                   torch.zeros(L, S, 64), False)
+    else:
+        ws = _gru_weights(L, cin, seed=9)
+        x = torch.randn(S, T, cin)
+        xr = x.clone().requires_grad_(True)
+        wr = [w.clone().requires_grad_(True) for w in ws]
+        out_ref = _gru_oracle(xr, wr, L, False)
+    (out_ref.float() ** 2).sum().backward()
+
+    xg = x.half().cuda().requires_grad_(True)
+    wg = [w.half().cuda().requires_grad_(True) for w in ws]
+    out = FusedLSTMFn.apply(xg, cell, False, True, *wg)
+    (out.float() ** 2).sum().backward()
+
+    def relerr(a, b):
+        return ((a.float().cpu() - b).abs().max() / (b.abs().max() + 1e-6)).item()
+
+    assert relerr(out.detach(), out_ref.detach()) < 0.05
+    assert relerr(xg.grad, xr.grad) < 0.1
+    for i, (a, b) in enumerate(zip(wg, wr)):
+        assert relerr(a.grad, b.grad) < 0.1, f"weight {i}"
