@@ -14,7 +14,6 @@ reference_impl.py (tests/test_gpu_kernels.py).
 """
 from __future__ import annotations
 
-from typing import List, Optional
 
 import torch
 

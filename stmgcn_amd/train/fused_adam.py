@@ -18,7 +18,7 @@ extension and a CUDA device).
 """
 from __future__ import annotations
 
-from typing import Iterable, List, Optional
+from typing import Iterable, List
 
 import torch
 

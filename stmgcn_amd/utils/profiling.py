@@ -17,7 +17,7 @@ from __future__ import annotations
 
 import json
 import time
-from typing import List, Optional
+from typing import List
 
 import torch
 
