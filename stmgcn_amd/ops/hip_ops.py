@@ -14,6 +14,7 @@ reference_impl.py (tests/test_gpu_kernels.py).
 """
 from __future__ import annotations
 
+from typing import Optional
 
 import torch
 
