@@ -158,7 +158,7 @@ extern "C" void stmgcn_atb_wgrad(void* stream, int dtype, const void* A,
                                  const void* B, float* C, float* db, long rows,
                                  int M, int N);
 
-static constexpr int kSeqTile = 64;
+static constexpr int kSeqTile = 32;  // MUST match SEQ_TILE in fused_rnn.hip
 static constexpr int kH = 64;
 
 std::vector<at::Tensor> lstm_fwd(at::Tensor x, std::vector<at::Tensor> w_ih,
