@@ -33,6 +33,11 @@ from stmgcn_amd.parallel import GradReducer, init_distributed, cleanup_distribut
 
 DTYPES = {"fp32": torch.float32, "bf16": torch.bfloat16, "fp16": torch.float16}
 
+# Self-established comparison floors: the reference publishes
+# no numbers, so the floor is the stock-PyTorch implementation (--impl torch)
+# measured on the same config / 1x MI355X. samples/s at N=1 GPU.
+BASELINE_FLOOR = {"bench-1024": 1198.0}
+
 
 def main():
     p = argparse.ArgumentParser()
@@ -186,6 +191,9 @@ def main():
 
     n_gpus = world if use_gpu else world  # ranks == GPUs in the driver launch
     samples_per_sec = n_gpus * B * args.steps / elapsed
+    # weak scaling: the floor scales with n_gpus for a whole-job ratio
+    floor = BASELINE_FLOOR.get(args.preset) if use_gpu else None
+    vs_baseline = (samples_per_sec / (floor * n_gpus)) if floor else None
     if rank == 0:
         rec = {
             "metric": "train_samples_per_sec",
@@ -197,7 +205,7 @@ def main():
             "ms_per_step": elapsed / args.steps * 1e3,
             "higher_is_better": True,
             "scaling": "weak",
-            "vs_baseline": None,
+            "vs_baseline": vs_baseline,
             "dtype": cfg.dtype if use_gpu else "fp32",
             "data": "synthetic",
             "config": {
