@@ -55,6 +55,8 @@ def main():
     p.add_argument("--sparse", action="store_true", help="CSR supports (HIP recurrence path)")
     p.add_argument("--model-dir", type=str, default="./output")
     p.add_argument("--metrics", type=str, default=None, help="JSONL metrics path")
+    p.add_argument("--graph", action="store_true",
+                   help="capture full-size train steps in a hipGraph (HIP path)")
     args = p.parse_args()
 
     cfg = PRESETS[args.preset]
@@ -128,7 +130,8 @@ def main():
     trainer = ModelTrainer(model=model, loss=loss, optimizer=opt_cls,
                            lr=cfg.lr, wd=cfg.weight_decay, n_epochs=cfg.n_epochs,
                            grad_reducer=reducer, rank=rank, world_size=world,
-                           metrics_path=args.metrics)
+                           metrics_path=args.metrics,
+                           use_graph=args.graph and hip_path)
 
     os.makedirs(args.model_dir, exist_ok=True)
     trainer.train(data_loader=loaders, sta_adj_list=sta_adj_list,

@@ -33,6 +33,8 @@ def main():
     p.add_argument("--nodes", type=int, default=None)
     p.add_argument("--days", type=int, default=21, help="synthetic dataset length")
     p.add_argument("--seed", type=int, default=7)
+    p.add_argument("--graph", action="store_true",
+                   help="hipGraph-captured train steps (HIP path)")
     args = p.parse_args()
     os.environ["STMGCN_IMPL"] = args.impl
 
@@ -90,7 +92,8 @@ def main():
     with tempfile.TemporaryDirectory() as model_dir:
         trainer = ModelTrainer(model=model, loss=loss, optimizer=opt_cls,
                                lr=cfg.lr, wd=cfg.weight_decay,
-                               n_epochs=cfg.n_epochs)
+                               n_epochs=cfg.n_epochs,
+                               use_graph=args.graph and hip_path)
         trainer.train(data_loader=loaders, sta_adj_list=adjs,
                       modes=["train", "validate"], model_dir=model_dir,
                       early_stopper=cfg.early_stop_patience)

@@ -42,7 +42,8 @@ class ModelTrainer:
 
     def __init__(self, model: nn.Module, loss: nn.Module, optimizer, lr: float,
                  wd: float, n_epochs: int, grad_reducer=None, rank: int = 0,
-                 world_size: int = 1, metrics_path: Optional[str] = None):
+                 world_size: int = 1, metrics_path: Optional[str] = None,
+                 use_graph: bool = False):
         self.model = model
         self.model_name = model.__class__.__name__
         if self.model_name not in self.SUPPORTED_MODELS:
@@ -54,6 +55,13 @@ class ModelTrainer:
         self.rank, self.world = rank, world_size
         self.metrics_path = metrics_path
         self._metrics_f = None
+        # whole-step hipGraph capture for full-size train batches (the
+        # bench-grade path: fwd+loss+bwd(+all-reduce)+Adam as one replay);
+        # ragged last batches and eval run eagerly. Requires the FusedAdam
+        # path (device-side bias correction) on GPU.
+        self.use_graph = use_graph and torch.cuda.is_available()
+        self._graph = None            # (graph, x_static, y_static, loss_static)
+        self._graph_shape = None
 
     # ------------------------------------------------------------------ utils
     def _log(self, *args):
@@ -83,6 +91,63 @@ class ModelTrainer:
     def _forward(self, x, sta_adj_list):
         return self.model(obs_seq=x, sta_adj_list=sta_adj_list)
 
+    def _train_step_eager(self, x, y_true, sta_adj_list):
+        if self.grad_reducer is not None:
+            self.grad_reducer.zero_grad()
+        elif hasattr(self.optimizer, "reduce"):
+            self.optimizer.zero_grad()           # FusedAdam arena
+        else:
+            self.optimizer.zero_grad(set_to_none=True)
+        y_pred = self._forward(x, sta_adj_list)
+        loss = self.criterion(y_pred, y_true)
+        loss.backward()
+        if self.grad_reducer is not None:
+            self.grad_reducer.reduce()
+        elif hasattr(self.optimizer, "reduce") and self.world > 1:
+            self.optimizer.reduce()              # one flat RCCL all-reduce
+        self.optimizer.step()
+        return loss
+
+    def _train_step(self, x, y_true, sta_adj_list):
+        """Replay the captured whole-step graph for full-size batches;
+        capture lazily on second sighting of the shape (after one eager
+        warmup); anything else runs eagerly."""
+        if not self.use_graph:
+            return self._train_step_eager(x, y_true, sta_adj_list)
+        shape = (tuple(x.shape), tuple(y_true.shape))
+        if self._graph is not None and self._graph_shape == shape:
+            g, xs, ys, ls = self._graph
+            xs.copy_(x)
+            ys.copy_(y_true)
+            g.replay()
+            return ls
+        if self._graph_shape != shape:           # first sighting: eager warmup
+            self._graph_shape = shape
+            self._graph = None
+            return self._train_step_eager(x, y_true, sta_adj_list)
+        # second sighting of the same shape: capture
+        try:
+            xs, ys = x.clone(), y_true.clone()
+            side = torch.cuda.Stream()
+            side.wait_stream(torch.cuda.current_stream())
+            with torch.cuda.stream(side):
+                self._train_step_eager(xs, ys, sta_adj_list)
+            torch.cuda.current_stream().wait_stream(side)
+            torch.cuda.synchronize()
+            g = torch.cuda.CUDAGraph()
+            with torch.cuda.graph(g):
+                ls = self._train_step_eager(xs, ys, sta_adj_list)
+            torch.cuda.synchronize()
+            self._graph = (g, xs, ys, ls)
+            xs.copy_(x)
+            ys.copy_(y_true)
+            g.replay()
+            return ls
+        except Exception as e:
+            self._log(f"# hipGraph capture unavailable, eager fallback: {e}")
+            self.use_graph = False
+            return self._train_step_eager(x, y_true, sta_adj_list)
+
     def _ckpt_path(self, model_dir: str) -> str:
         return os.path.join(model_dir, f"{self.model_name}_best_model.pkl")
 
@@ -102,22 +167,12 @@ class ModelTrainer:
                 self.model.train() if mode == "train" else self.model.eval()
                 step = 0
                 for x, y_true in data_loader[mode]:
-                    with torch.set_grad_enabled(mode == "train"):
-                        y_pred = self._forward(x, sta_adj_list)
-                        loss = self.criterion(y_pred, y_true)
-                        if mode == "train":
-                            if self.grad_reducer is not None:
-                                self.grad_reducer.zero_grad()
-                            elif hasattr(self.optimizer, "reduce"):
-                                self.optimizer.zero_grad()   # FusedAdam arena
-                            else:
-                                self.optimizer.zero_grad(set_to_none=True)
-                            loss.backward()
-                            if self.grad_reducer is not None:
-                                self.grad_reducer.reduce()
-                            elif hasattr(self.optimizer, "reduce") and self.world > 1:
-                                self.optimizer.reduce()      # one flat RCCL all-reduce
-                            self.optimizer.step()
+                    if mode == "train":
+                        loss = self._train_step(x, y_true, sta_adj_list)
+                    else:
+                        with torch.no_grad():
+                            loss = self.criterion(self._forward(x, sta_adj_list),
+                                                  y_true)
                     running_loss[mode] += float(loss.detach()) * y_true.shape[0]
                     step += y_true.shape[0]
                 if mode == "train":
