@@ -108,23 +108,26 @@ class ModelTrainer:
         self.optimizer.step()
         return loss
 
-    def _train_step(self, x, y_true, sta_adj_list):
-        """Replay the captured whole-step graph for full-size batches;
-        capture lazily on second sighting of the shape (after one eager
-        warmup); anything else runs eagerly."""
+    def _train_step(self, x, y_true, sta_adj_list) -> float:
+        """Returns the step loss as a FLOAT — never a live tensor: a loss
+        held by the caller across iterations keeps the autograd graph (and
+        its default-stream AccumulateGrad nodes) alive, which breaks hipGraph
+        capture of the next step. Replays the captured whole-step graph for
+        full-size batches; captures lazily on the second sighting of a shape
+        (after one eager warmup); anything else runs eagerly."""
         if not self.use_graph:
-            return self._train_step_eager(x, y_true, sta_adj_list)
+            return float(self._train_step_eager(x, y_true, sta_adj_list).detach())
         shape = (tuple(x.shape), tuple(y_true.shape))
         if self._graph is not None and self._graph_shape == shape:
             g, xs, ys, ls = self._graph
             xs.copy_(x)
             ys.copy_(y_true)
             g.replay()
-            return ls
+            return float(ls.detach())
         if self._graph_shape != shape:           # first sighting: eager warmup
             self._graph_shape = shape
             self._graph = None
-            return self._train_step_eager(x, y_true, sta_adj_list)
+            return float(self._train_step_eager(x, y_true, sta_adj_list).detach())
         # second sighting of the same shape: capture
         try:
             xs, ys = x.clone(), y_true.clone()
@@ -142,11 +145,11 @@ class ModelTrainer:
             xs.copy_(x)
             ys.copy_(y_true)
             g.replay()
-            return ls
+            return float(ls.detach())
         except Exception as e:
             self._log(f"# hipGraph capture unavailable, eager fallback: {e}")
             self.use_graph = False
-            return self._train_step_eager(x, y_true, sta_adj_list)
+            return float(self._train_step_eager(x, y_true, sta_adj_list).detach())
 
     def _ckpt_path(self, model_dir: str) -> str:
         return os.path.join(model_dir, f"{self.model_name}_best_model.pkl")
@@ -168,12 +171,12 @@ class ModelTrainer:
                 step = 0
                 for x, y_true in data_loader[mode]:
                     if mode == "train":
-                        loss = self._train_step(x, y_true, sta_adj_list)
+                        lval = self._train_step(x, y_true, sta_adj_list)
                     else:
                         with torch.no_grad():
-                            loss = self.criterion(self._forward(x, sta_adj_list),
-                                                  y_true)
-                    running_loss[mode] += float(loss.detach()) * y_true.shape[0]
+                            lval = float(self.criterion(
+                                self._forward(x, sta_adj_list), y_true).detach())
+                    running_loss[mode] += lval * y_true.shape[0]
                     step += y_true.shape[0]
                 if mode == "train":
                     n_train_samples = step * self.world

@@ -590,8 +590,7 @@ def test_trainer_hipgraph_capture_path():
                           lr=5e-3, wd=0.0, n_epochs=1, use_graph=graph)
         ls = []
         for step in range(12):
-            loss = tr._train_step(x[:8], y[:8], adjs)
-            ls.append(float(loss.detach()))
+            ls.append(tr._train_step(x[:8], y[:8], adjs))  # returns float
         losses[graph] = ls
         assert all(np.isfinite(v) for v in ls)
     assert tr._graph is not None, "graph was never captured"
