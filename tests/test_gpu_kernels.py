@@ -482,7 +482,7 @@ def test_fused_trainer_path_loss_decreases():
         loss = mse_loss(model(x, adjs), y)
         loss.backward()
         opt.step()
-        losses.append(float(loss))
+        losses.append(float(loss.detach()))
     assert all(torch.isfinite(p.float()).all() for p in model.parameters())
     assert losses[-1] < 0.5 * losses[0], f"no learning: {losses[0]} -> {losses[-1]}"
 
