@@ -57,6 +57,8 @@ def main():
     p.add_argument("--metrics", type=str, default=None, help="JSONL metrics path")
     p.add_argument("--graph", action="store_true",
                    help="capture full-size train steps in a hipGraph (HIP path)")
+    p.add_argument("--resume", action="store_true",
+                   help="resume from the best-val checkpoint in --model-dir")
     args = p.parse_args()
 
     cfg = PRESETS[args.preset]
@@ -134,9 +136,10 @@ def main():
                            use_graph=args.graph and hip_path)
 
     os.makedirs(args.model_dir, exist_ok=True)
+    start_epoch = trainer.resume(args.model_dir) if args.resume else 0
     trainer.train(data_loader=loaders, sta_adj_list=sta_adj_list,
                   modes=["train", "validate"], model_dir=args.model_dir,
-                  early_stopper=cfg.early_stop_patience)
+                  early_stopper=cfg.early_stop_patience, start_epoch=start_epoch)
     if rank == 0:
         print("Test: on Month", args.dates[2][:2], "Model", args.model_name)
         trainer.test(data_loader=loaders, sta_adj_list=sta_adj_list,

@@ -154,15 +154,39 @@ class ModelTrainer:
     def _ckpt_path(self, model_dir: str) -> str:
         return os.path.join(model_dir, f"{self.model_name}_best_model.pkl")
 
+    def _optim_path(self, model_dir: str) -> str:
+        # sidecar file: the reference-format best_model.pkl stays
+        # byte-compatible ({'epoch', 'state_dict'} only — SURVEY §5)
+        return os.path.join(model_dir, f"{self.model_name}_best_model.optim.pkl")
+
+    def resume(self, model_dir: str) -> int:
+        """Failure recovery: reload the best-val checkpoint (and
+        optimizer state if its sidecar exists) and return the epoch to
+        continue from. No-op (returns 0) when no checkpoint exists."""
+        path = self._ckpt_path(model_dir)
+        if not os.path.exists(path):
+            return 0
+        saved = torch.load(path, weights_only=False, map_location="cpu")
+        dev = next(self.model.parameters()).device
+        dt = next(self.model.parameters()).dtype
+        self.model.load_state_dict(
+            {k: v.to(device=dev, dtype=dt) for k, v in saved["state_dict"].items()})
+        opath = self._optim_path(model_dir)
+        if os.path.exists(opath):
+            self.optimizer.load_state_dict(
+                torch.load(opath, weights_only=False, map_location=dev))
+        self._log(f"Resumed from {path} (epoch {saved['epoch']})")
+        return int(saved["epoch"])
+
     # ------------------------------------------------------------------ train
     def train(self, data_loader: Dict, sta_adj_list: List, modes: List[str],
-              model_dir: str, early_stopper: int = 10):
+              model_dir: str, early_stopper: int = 10, start_epoch: int = 0):
         patience = early_stopper
-        checkpoint = {"epoch": 0, "state_dict": self.model.state_dict()}
+        checkpoint = {"epoch": start_epoch, "state_dict": self.model.state_dict()}
         val_loss = np.inf
         self._log("Training starts at: ", time.ctime())
 
-        for epoch in range(1, self.n_epochs + 1):
+        for epoch in range(start_epoch + 1, self.n_epochs + 1):
             running_loss = {mode: 0.0 for mode in modes}
             t_epoch = time.perf_counter()
             n_train_samples = 0
@@ -195,6 +219,8 @@ class ModelTrainer:
                         checkpoint.update(epoch=epoch, state_dict=self.model.state_dict())
                         if self.rank == 0:
                             torch.save(checkpoint, self._ckpt_path(model_dir))
+                            torch.save(self.optimizer.state_dict(),
+                                       self._optim_path(model_dir))
                         patience = early_stopper
                     else:
                         self._log(f"Epoch {epoch}, Val_loss does not improve from {val_loss:.5}.")

@@ -165,3 +165,51 @@ def test_main_cli_end_to_end(tmp_path):
     saved = torch.load(ckpt, weights_only=False)
     assert set(saved) == {"epoch", "state_dict"}
     assert (tmp_path / "metrics.jsonl").read_text().strip()
+
+
+def test_resume_from_best_checkpoint(tmp_path):
+    """Failure recovery: kill after 2 epochs, resume — weights
+    and optimizer state reload, training continues from the saved epoch,
+    and the reference checkpoint file format is unchanged (no extra keys)."""
+    import torch.optim as optim
+    from stmgcn_amd import PRESETS
+    from stmgcn_amd.graph import SupportGenerator
+    from stmgcn_amd.models import build_model
+    from stmgcn_amd.train import ModelTrainer
+
+    cfg = PRESETS["cpu-small"].replace(n_nodes=16, batch_size=8, seq_len=4,
+                                       lstm_hidden_dim=8, gcn_hidden_dim=8,
+                                       lstm_num_layers=1, m_graphs=1,
+                                       obs_len=[4, 0, 0])
+    torch.manual_seed(0)
+    gen = SupportGenerator(cfg.kernel_type, cfg.cheby_K, cfg.lambda_max_mode)
+    a = torch.rand(16, 16)
+    a = ((a + a.T) > 1.4).float()
+    a.fill_diagonal_(0)
+    adjs = [gen.process(a)]
+    x = torch.randn(24, cfg.seq_len, 16, 1)
+    y = torch.randn(24, 16, 1)
+    from stmgcn_amd.data.container import DeviceLoader
+    loaders = {m: DeviceLoader(x, y, 8) for m in ["train", "validate"]}
+
+    def make_trainer():
+        torch.manual_seed(1)
+        model = build_model(cfg)
+        return ModelTrainer(model=model, loss=torch.nn.MSELoss(),
+                            optimizer=optim.Adam, lr=1e-2, wd=0.0, n_epochs=2)
+
+    t1 = make_trainer()
+    t1.train(loaders, adjs, ["train", "validate"], str(tmp_path))
+    saved = torch.load(tmp_path / "ST_MGCN_best_model.pkl", weights_only=False)
+    assert set(saved) == {"epoch", "state_dict"}      # reference layout intact
+    assert (tmp_path / "ST_MGCN_best_model.optim.pkl").exists()
+
+    t2 = make_trainer()                                # fresh process stand-in
+    start = t2.resume(str(tmp_path))
+    assert start == saved["epoch"] > 0
+    for k, v in t2.model.state_dict().items():
+        torch.testing.assert_close(v, saved["state_dict"][k])
+    assert t2.optimizer.state_dict()["state"]          # optimizer moments loaded
+    t2.n_epochs = start + 1
+    t2.train(loaders, adjs, ["train", "validate"], str(tmp_path),
+             start_epoch=start)                        # continues, no crash
