@@ -213,3 +213,65 @@ def test_resume_from_best_checkpoint(tmp_path):
     t2.n_epochs = start + 1
     t2.train(loaders, adjs, ["train", "validate"], str(tmp_path),
              start_epoch=start)                        # continues, no crash
+
+
+def _tiny_setup(seed=0):
+    import torch.optim as optim
+    from stmgcn_amd import PRESETS
+    from stmgcn_amd.data.container import DeviceLoader
+    from stmgcn_amd.graph import SupportGenerator
+    from stmgcn_amd.models import build_model
+    from stmgcn_amd.train import ModelTrainer
+
+    cfg = PRESETS["cpu-small"].replace(n_nodes=12, batch_size=6, seq_len=4,
+                                       lstm_hidden_dim=4, gcn_hidden_dim=4,
+                                       lstm_num_layers=1, m_graphs=1,
+                                       obs_len=[4, 0, 0])
+    torch.manual_seed(seed)
+    gen = SupportGenerator(cfg.kernel_type, cfg.cheby_K, cfg.lambda_max_mode)
+    a = torch.rand(12, 12)
+    a = ((a + a.T) > 1.4).float()
+    a.fill_diagonal_(0)
+    adjs = [gen.process(a)]
+    x = torch.randn(12, cfg.seq_len, 12, 1)
+    y = torch.randn(12, 12, 1)
+    loaders = {m: DeviceLoader(x, y, 6) for m in ["train", "validate"]}
+
+    def make(n_epochs):
+        torch.manual_seed(seed + 1)
+        model = build_model(cfg)
+        return ModelTrainer(model=model, loss=torch.nn.MSELoss(),
+                            optimizer=optim.Adam, lr=1e-3, wd=0.0,
+                            n_epochs=n_epochs)
+    return make, loaders, adjs
+
+
+def test_early_stopping_fires(tmp_path, capsys):
+    """Patience exhausts -> 'Early stopping at epoch N..' and train returns
+    before n_epochs (reference Model_Trainer.py:55-60)."""
+    make, loaders, adjs = _tiny_setup()
+    tr = make(n_epochs=50)
+    tr.train(loaders, adjs, ["train", "validate"], str(tmp_path),
+             early_stopper=2)
+    out = capsys.readouterr().out
+    assert "Early stopping at epoch" in out
+    # stopped well before 50 epochs
+    stopped = int(out.split("Early stopping at epoch")[1].split("..")[0])
+    assert stopped < 50
+
+
+def test_final_save_is_best_not_last(tmp_path):
+    """The end-of-training save re-saves the BEST checkpoint dict, not the
+    last epoch's weights (reference quirk 10 semantics)."""
+    make, loaders, adjs = _tiny_setup(seed=3)
+    tr = make(n_epochs=3)
+    tr.train(loaders, adjs, ["train", "validate"], str(tmp_path),
+             early_stopper=50)
+    saved = torch.load(tmp_path / "ST_MGCN_best_model.pkl", weights_only=False)
+    # the saved epoch is the best-val epoch; weights differ from the live
+    # (last-epoch) model unless the best WAS the last epoch
+    assert 1 <= saved["epoch"] <= 3
+    if saved["epoch"] != 3:
+        live = tr.model.state_dict()
+        assert any(not torch.equal(v, live[k])
+                   for k, v in saved["state_dict"].items())
