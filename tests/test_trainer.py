@@ -237,11 +237,11 @@ def _tiny_setup(seed=0):
     y = torch.randn(12, 12, 1)
     loaders = {m: DeviceLoader(x, y, 6) for m in ["train", "validate"]}
 
-    def make(n_epochs):
+    def make(n_epochs, lr=1e-3):
         torch.manual_seed(seed + 1)
         model = build_model(cfg)
         return ModelTrainer(model=model, loss=torch.nn.MSELoss(),
-                            optimizer=optim.Adam, lr=1e-3, wd=0.0,
+                            optimizer=optim.Adam, lr=lr, wd=0.0,
                             n_epochs=n_epochs)
     return make, loaders, adjs
 
@@ -250,7 +250,7 @@ def test_early_stopping_fires(tmp_path, capsys):
     """Patience exhausts -> 'Early stopping at epoch N..' and train returns
     before n_epochs (reference Model_Trainer.py:55-60)."""
     make, loaders, adjs = _tiny_setup()
-    tr = make(n_epochs=50)
+    tr = make(n_epochs=50, lr=1.0)   # divergent lr -> val loss worsens
     tr.train(loaders, adjs, ["train", "validate"], str(tmp_path),
              early_stopper=2)
     out = capsys.readouterr().out
