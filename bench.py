@@ -53,6 +53,9 @@ def main():
                    help="write a torch profiler trace to this path")
     p.add_argument("--graph", type=str, default="auto", choices=["auto", "on", "off"],
                    help="capture the whole train step in a hipGraph and replay it")
+    p.add_argument("--mode", type=str, default="train", choices=["train", "infer"],
+                   help="train: fwd+loss+bwd+opt (the driver contract); "
+                        "infer: eval-mode forward only (the serving path)")
     args = p.parse_args()
 
     if args.impl:
@@ -111,7 +114,14 @@ def main():
                          weight_decay=cfg.weight_decay,
                          capturable=want_graph, foreach=True)
 
+    infer = args.mode == "infer"
+    if infer:
+        model.eval()
+
     def step():
+        if infer:
+            with torch.no_grad():
+                return model(x, adjs)
         if reducer is not None:
             reducer.zero_grad()
         else:
@@ -196,7 +206,7 @@ def main():
     vs_baseline = (samples_per_sec / (floor * n_gpus)) if floor else None
     if rank == 0:
         rec = {
-            "metric": "train_samples_per_sec",
+            "metric": "infer_samples_per_sec" if infer else "train_samples_per_sec",
             "value": samples_per_sec,
             "unit": "samples/s",
             "n_gpus": n_gpus,
