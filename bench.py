@@ -202,7 +202,8 @@ def main():
     n_gpus = world if use_gpu else world  # ranks == GPUs in the driver launch
     samples_per_sec = n_gpus * B * args.steps / elapsed
     # weak scaling: the floor scales with n_gpus for a whole-job ratio
-    floor = BASELINE_FLOOR.get(args.preset) if use_gpu else None
+    # (train-mode floors only — no published/measured infer floor)
+    floor = BASELINE_FLOOR.get(args.preset) if (use_gpu and not infer) else None
     vs_baseline = (samples_per_sec / (floor * n_gpus)) if floor else None
     if rank == 0:
         rec = {
