@@ -1,1 +1,2 @@
-from .misc import seed_everything, StepTimer  # noqa: F401
+from .misc import seed_everything  # noqa: F401
+from .profiling import StepTimer  # noqa: F401
