@@ -130,8 +130,14 @@ lstm_fwd_kernel(const T* __restrict__ x,    // (S, Tst, C_in)
   const int lgrp = lane >> 4;
   const long S_pad = (long)gridDim.x * ST;
 
+  // CIN1: ping/pong h buffers by LAYER parity — the writer never aliases
+  // the prev-layer slots being read, so the pre-write WAR barrier drops
+  // (one barrier per (layer,t) stage instead of two). Dense-input variants
+  // keep the single in-place buffer (a third T*SLOT buffer would push LDS
+  // past 2-WG occupancy).
   char* hseq = lds;                                   // Tst * SLOT bytes
-  char* xbuf = lds + Tst * SLOT;
+  char* hpong = lds + (CIN1 ? Tst * SLOT : 0);        // CIN1 only
+  char* xbuf = lds + (CIN1 ? 2 : 1) * (long)Tst * SLOT;
 
   // ---- stage input x into LDS --------------------------------------------
   if (CIN1) {
@@ -160,6 +166,9 @@ lstm_fwd_kernel(const T* __restrict__ x,    // (S, Tst, C_in)
 
   for (int layer = 0; layer < L; ++layer) {
     const int cin = (layer == 0) ? (CIN1 ? 1 : RNN_H) : RNN_H;
+    // this layer's write buffer / previous layer's read buffer
+    char* wrbuf = CIN1 ? ((layer & 1) ? hseq : hpong) : hseq;
+    char* rdbuf = CIN1 ? ((layer & 1) ? hpong : hseq) : hseq;
     const T* Whh = (const T*)ptrs.w_hh[layer];
     const T* Wih = (const T*)ptrs.w_ih[layer];
     const T* bih = (const T*)ptrs.b_ih[layer];
@@ -204,9 +213,9 @@ lstm_fwd_kernel(const T* __restrict__ x,    // (S, Tst, C_in)
         #pragma unroll
       for (int q = 0; q < 4; ++q) acc[m][q] = f32x4{0.f, 0.f, 0.f, 0.f};
 
-      // recurrent term: h_{t-1} from hseq slot t-1 (zero at t == 0)
+      // recurrent term: h_{t-1} from this layer's write buffer (zero at t==0)
       if (t > 0) {
-        char* slot = hseq + (t - 1) * SLOT;
+        char* slot = wrbuf + (t - 1) * SLOT;
         #pragma unroll
       for (int m = 0; m < MT; ++m) {
           const int row = 16 * m + l16;
@@ -219,9 +228,10 @@ lstm_fwd_kernel(const T* __restrict__ x,    // (S, Tst, C_in)
           }
         }
       }
-      // input term
+      // input term: previous layer's h (dense layer 0: the staged x)
       if (!CIN1 || layer > 0) {
-        char* src = (layer == 0) ? (xbuf + t * SLOT) : (hseq + t * SLOT);
+        char* src = (!CIN1 && layer == 0) ? (xbuf + t * SLOT)
+                                          : (rdbuf + t * SLOT);
         #pragma unroll
       for (int m = 0; m < MT; ++m) {
           const int row = 16 * m + l16;
@@ -285,10 +295,11 @@ lstm_fwd_kernel(const T* __restrict__ x,    // (S, Tst, C_in)
         }
       }
 
-      // wait: all waves finished reading slot t (input term) before overwrite
-      __syncthreads();
+      // dense in-place variant: all waves must finish reading slot t
+      // (input term) before it is overwritten; ping/pong needs no WAR wait
+      if (!CIN1) __syncthreads();
       {
-        char* slot = hseq + t * SLOT;
+        char* slot = wrbuf + t * SLOT;
         #pragma unroll
       for (int m = 0; m < MT; ++m)
           #pragma unroll
@@ -323,10 +334,10 @@ lstm_fwd_kernel(const T* __restrict__ x,    // (S, Tst, C_in)
           *(ulong1*)(cp + (m * 64 + lane) * 4) = *(ulong1*)c4;
         }
         // hseq natural layout copy of slot t (also next layer's input source)
-        // (no trailing barrier: slot t's next write is the next layer's
-        // stage t, separated by that stage's pre-write barrier)
+        // (no trailing barrier: the next write to this address is at least
+        // one barrier away in both variants)
         T* hp = hseq_g + base * (S_pad * RNN_H) + (long)(s0)*RNN_H;
-        char* slot = hseq + t * SLOT;
+        char* slot = wrbuf + t * SLOT;
         for (int i = threadIdx.x; i < ST * 8; i += 256) {
           const int c8 = i & 7, s = i >> 3;
           *(frag*)&hp[s * RNN_H + c8 * 8] = *(frag*)&slot[lds_swz(s, c8 * 16)];
@@ -335,16 +346,17 @@ lstm_fwd_kernel(const T* __restrict__ x,    // (S, Tst, C_in)
     }
   }
 
-  // ---- output ------------------------------------------------------------
+  // ---- output (from the LAST layer's write buffer) -----------------------
+  char* lastbuf = CIN1 ? (((L - 1) & 1) ? hseq : hpong) : hseq;
   if (ret_seq) {
     for (int i = threadIdx.x; i < ST * Tst * 8; i += 256) {
       const int c8 = i & 7, s = (i >> 3) & (ST - 1), t = (i >> 3) / ST;
       if (s0 + s < S)
         *(frag*)&out[((long)(s0 + s) * Tst + t) * RNN_H + c8 * 8] =
-            *(frag*)&hseq[t * SLOT + lds_swz(s, c8 * 16)];
+            *(frag*)&lastbuf[t * SLOT + lds_swz(s, c8 * 16)];
     }
   } else {
-    char* slot = hseq + (Tst - 1) * SLOT;
+    char* slot = lastbuf + (Tst - 1) * SLOT;
     for (int i = threadIdx.x; i < ST * 8; i += 256) {
       const int c8 = i & 7, s = i >> 3;
       if (s0 + s < S)
@@ -362,7 +374,9 @@ void launch_fwd(hipStream_t stream, const void* x, void* out, void* hseq_g,
   const int nblk = (S + ST - 1) / ST;
   const bool cin1 = (cin == 1);
   const size_t slot = (size_t)ST * 128;
-  const size_t lds_bytes = (size_t)Tst * slot + (cin1 ? Tst * ST * sizeof(T) : (size_t)Tst * slot);
+  const size_t lds_bytes = cin1
+      ? 2 * (size_t)Tst * slot + Tst * ST * sizeof(T)   // ping/pong + x scalars
+      : 2 * (size_t)Tst * slot;                         // in-place h + dense x
   auto go = [&](auto kern) {
     hipLaunchKernelGGL(kern, dim3(nblk), dim3(256), lds_bytes, stream,
                        (const T*)x, (T*)out, (T*)hseq_g, (T*)cseq_g,
@@ -447,8 +461,12 @@ lstm_bwd_kernel(const T* __restrict__ dout,     // (S,H) or (S,Tst,H)
   const int hch = 16 * wv + l16;
 
   char* dh_lds = lds;                         // [Tst][ST][64] fwd-slot layout
-  char* dA_lds = lds + Tst * SLOT;            // [ST][512B] swizzled
-  float* red = (float*)(dA_lds + ST * 512);   // [4][ST] cross-wave scratch
+  // dA tiles ping/pong by timestep parity: the pointwise writes of step t
+  // never alias the buffer step t+1's GEMMs/stream read, so the loop-top
+  // WAR barrier drops (one barrier per stage; bwd was 70% wave-parked)
+  char* dA_buf0 = lds + Tst * SLOT;           // [ST][512B] swizzled
+  char* dA_buf1 = dA_buf0 + ST * 512;
+  float* red = (float*)(dA_buf1 + ST * 512);  // [4][ST] cross-wave scratch
 
   for (int layer = L - 1; layer >= 0; --layer) {
     const bool l0cin1 = CIN1 && layer == 0;
@@ -478,11 +496,12 @@ lstm_bwd_kernel(const T* __restrict__ dout,     // (S,H) or (S,Tst,H)
       gcur[m][0] = *(((const frag*)(g_at(Tst - 1) + (m * 64 + lane) * 16)) + 0);
       gcur[m][1] = *(((const frag*)(g_at(Tst - 1) + (m * 64 + lane) * 16)) + 1);
     }
+    // layer boundary: for odd Tst the first stage reuses the buffer parity
+    // of the previous layer's last stage — its readers must be done
+    __syncthreads();
 
     for (int t = Tst - 1; t >= 0; --t) {
-      // WAR: previous step's reads of dA_lds must be complete
-      __syncthreads();
-
+      char* dA_lds = (t & 1) ? dA_buf1 : dA_buf0;
       const long base = lay_base + t;
 
       float wih0[4];
@@ -669,7 +688,7 @@ void launch_bwd(hipStream_t stream, const void* dout, const void* x,
                 int ret_seq, int gru) {
   constexpr int ST = SEQ_TILE;
   const int nblk = (S + ST - 1) / ST;
-  const size_t lds_bytes = (size_t)Tst * ST * 128 + ST * 512 + 4 * ST * sizeof(float);
+  const size_t lds_bytes = (size_t)Tst * ST * 128 + 2 * ST * 512 + 4 * ST * sizeof(float);
   auto go = [&](auto kern) {
     hipLaunchKernelGGL(kern, dim3(nblk), dim3(256), lds_bytes, stream,
                        (const T*)dout, (const T*)x, (const T*)cseq_g,
