@@ -75,8 +75,13 @@ def main():
 
     env = init_distributed()
     rank, world = env["rank"], env["world_size"]
-    device = torch.device(f"cuda:{env['local_rank']}") if str(args.device).startswith("cuda") \
-        else torch.device("cpu")
+    if not str(args.device).startswith("cuda"):
+        device = torch.device("cpu")
+    elif world > 1 or "LOCAL_RANK" in os.environ:
+        device = torch.device(f"cuda:{env['local_rank']}")  # one rank per GPU
+    else:
+        # single-process: honor -device cuda:N (reference Main.py:22-23)
+        device = torch.device(args.device)
     dtype = DTYPES[cfg.dtype]
     if device.type == "cpu":
         dtype = torch.float32  # CPU oracle path runs fp32

@@ -192,6 +192,14 @@ def main():
         if rank == 0:
             prof_ctx.export_chrome_trace(args.profile_trace)
 
+    # impl truthfulness: if any op fell back to torch under
+    # STMGCN_ALLOW_FALLBACK=1, the JSON must not claim a pure-hip run
+    impl = os.environ.get("STMGCN_IMPL", "hip") if use_gpu else "torch-cpu"
+    if impl == "hip":
+        from stmgcn_amd.ops import hip_ops
+        if hip_ops.fallback_count() > 0:
+            impl = "hip+fallback"
+
     # max over ranks
     if world > 1:
         t = torch.tensor([elapsed], dtype=torch.float64,
@@ -227,7 +235,7 @@ def main():
                 "global_batch": B * n_gpus,
                 "parallelism": f"dp{n_gpus}",
                 "hipgraph": graph_ok,
-                "impl": os.environ.get("STMGCN_IMPL", "hip") if use_gpu else "torch-cpu",
+                "impl": impl,
             },
         }
         print(json.dumps(rec))

@@ -258,7 +258,9 @@ std::vector<at::Tensor> atb_wgrad(at::Tensor A, at::Tensor B, bool want_db) {
   A = A.contiguous(); B = B.contiguous();
   const long rows = A.size(0);
   const int M = A.size(1), N = B.size(1);
-  TORCH_CHECK(B.size(0) == rows && M <= 256 && N <= 64);
+  // M/N multiples of 8: the kernel reads 16-byte fragments at column
+  // offsets cb*8; a ragged tail row would read past the end of A/B.
+  TORCH_CHECK(B.size(0) == rows && M <= 256 && N <= 64 && M % 8 == 0 && N % 8 == 0);
   auto fopt = A.options().dtype(at::kFloat);
   auto C = at::zeros({M, N}, fopt);
   auto db = want_db ? at::zeros({N}, fopt) : at::Tensor();

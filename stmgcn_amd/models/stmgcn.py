@@ -51,6 +51,25 @@ def _branch_streams(n: int) -> List[torch.cuda.Stream]:
     return _STREAMS[:n]
 
 
+def _norm_activation(act):
+    """Accept the reference's nn-module wiring (Main.py:64 passes
+    gconv_activation=nn.ReLU) alongside the string form used here."""
+    if act is None or isinstance(act, str):
+        return act
+    if act is nn.ReLU or isinstance(act, nn.ReLU):
+        return "relu"
+    raise ValueError(
+        f"unsupported gconv activation {act!r}: pass 'relu', None, or nn.ReLU")
+
+
+def _record_stream(t: torch.Tensor, stream: torch.cuda.Stream) -> None:
+    # During hipGraph capture the graph's private memory pool keeps
+    # allocations stable, and some torch/ROCm builds reject record_stream
+    # inside capture — skip it there so --graph keeps the multi-stream path.
+    if not torch.cuda.is_current_stream_capturing():
+        t.record_stream(stream)
+
+
 class GCN(nn.Module):
     """K-support graph convolution op (reference GCN.py:7-46).
 
@@ -63,7 +82,7 @@ class GCN(nn.Module):
         self.K = K
         self.input_dim = input_dim
         self.hidden_dim = hidden_dim
-        self.activation = activation
+        self.activation = _norm_activation(activation)
         self.W = nn.Parameter(torch.empty(K * input_dim, hidden_dim))
         nn.init.xavier_normal_(self.W)
         self.b = nn.Parameter(torch.zeros(hidden_dim)) if bias else None
@@ -216,13 +235,13 @@ class ST_MGCN(nn.Module):
             for m in range(self.M):
                 streams[m].wait_event(fork)
                 with torch.cuda.stream(streams[m]):
-                    obs_seq.record_stream(streams[m])
+                    _record_stream(obs_seq, streams[m])
                     h = self.rnn_list[m](sta_adj_list[m], obs_seq)
                     f = self.gcn_list[m](sta_adj_list[m], h)
                 feat_list.append(f)
             for m in range(self.M):
                 main.wait_stream(streams[m])
-                feat_list[m].record_stream(main)
+                _record_stream(feat_list[m], main)
         else:
             for m in range(self.M):
                 h = self.rnn_list[m](sta_adj_list[m], obs_seq)     # (B,N,H)
@@ -271,11 +290,11 @@ class STMGCNBlock(nn.Module):
             for m in range(self.M):
                 streams[m].wait_event(fork)
                 with torch.cuda.stream(streams[m]):
-                    x.record_stream(streams[m])
+                    _record_stream(x, streams[m])
                     outs.append(branch(m))
             for m in range(self.M):
                 main.wait_stream(streams[m])
-                outs[m].record_stream(main)
+                _record_stream(outs[m], main)
         else:
             outs = [branch(m) for m in range(self.M)]
         out = outs[0]

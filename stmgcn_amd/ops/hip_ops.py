@@ -18,9 +18,39 @@ from typing import Optional
 
 import torch
 
+import os
+
 from . import reference_impl as ref
 from .functional import require_hip
 from ..graph.preprocess import CSRSupport
+
+# ---------------------------------------------------------------------------
+# Fallback policy: the HIP kernels serve fixed shapes (H=64, T<=16, ...).
+# Off-shape on the GPU hot path is a HARD ERROR unless the caller explicitly
+# sets STMGCN_ALLOW_FALLBACK=1 — a silent torch fallback would bench stock
+# PyTorch while reporting impl="hip" (fail-loud principle, functional.py).
+# Every allowed fallback is counted so bench.py can surface impl="hip+fallback".
+_fallback_count = 0
+
+
+def fallback_count() -> int:
+    return _fallback_count
+
+
+def reset_fallback_count() -> None:
+    global _fallback_count
+    _fallback_count = 0
+
+
+def _fallback(what: str) -> None:
+    global _fallback_count
+    if os.environ.get("STMGCN_ALLOW_FALLBACK", "0") != "1":
+        raise RuntimeError(
+            f"HIP kernel path cannot serve {what}; refusing a silent torch "
+            "fallback on the GPU hot path. Set STMGCN_ALLOW_FALLBACK=1 to run "
+            "the torch op anyway (reported as impl=hip+fallback), or "
+            "STMGCN_IMPL=torch for the full stock-PyTorch path.")
+    _fallback_count += 1
 
 
 class ChebGconvFn(torch.autograd.Function):
@@ -72,8 +102,11 @@ class ChebGconvFn(torch.autograd.Function):
         Cout = dz.shape[-1]
         # tall-skinny reduction GEMM kernel (wgrad.hip): dW = feat^T dZ and
         # db = colsum(dZ) in one launch, fp32 accumulate. (bf16/f16 only —
-        # the MFMA tile is 16x16x32_bf16/f16; fp32 parity runs rocBLAS.)
-        if dz.dtype in (torch.bfloat16, torch.float16) and KC <= 256 and Cout <= 64:
+        # the MFMA tile is 16x16x32_bf16/f16; fp32 parity runs rocBLAS.
+        # M/N must be multiples of 8: the kernel issues 16-byte fragment
+        # loads at column offsets cb*8 — a ragged tail would read OOB.)
+        if (dz.dtype in (torch.bfloat16, torch.float16) and KC <= 256
+                and Cout <= 64 and KC % 8 == 0 and Cout % 8 == 0):
             outs = C.atb_wgrad(feat.reshape(-1, KC), dz.reshape(-1, Cout), ctx.has_b)
             dW = outs[0].to(W.dtype)
             db = outs[1].to(W.dtype) if ctx.has_b else None
@@ -89,6 +122,7 @@ class ChebGconvFn(torch.autograd.Function):
 def contextual_gate_hip(obs_seq, gconv_out, fc_weight, fc_bias):
     if obs_seq.shape[1] <= 16:
         return GateFn.apply(obs_seq, gconv_out, fc_weight, fc_bias)
+    _fallback(f"contextual gate with T={obs_seq.shape[1]} > 16")
     return ref.contextual_gate(obs_seq, gconv_out, fc_weight, fc_bias)
 
 
@@ -213,6 +247,10 @@ class FusedRNNFn:
             training = torch.is_grad_enabled() and (
                 x.requires_grad or any(w.requires_grad for w in weights))
             return FusedLSTMFn.apply(x, cell, return_sequences, training, *weights)
+        _fallback(
+            f"fused {cell} with dtype={x.dtype}, C_in={x.shape[-1]}, "
+            f"T={x.shape[1]}, hidden={tuple(weights[1].shape)} "
+            "(served: bf16/f16, C_in in {1,64}, T<=16, H=64)")
         return _vf_rnn(cell, x, h0, c0, return_sequences, weights)
 
 
@@ -220,6 +258,9 @@ def branch_fuse_head_hip(branch_feats, fc_weight, fc_bias):
     if (fc_weight.shape[0] == 1 and fc_weight.shape[1] <= 64
             and len(branch_feats) <= 3):
         return HeadFn.apply(fc_weight, fc_bias, *branch_feats)
+    _fallback(
+        f"fused head with C_out={fc_weight.shape[0]}, G={fc_weight.shape[1]}, "
+        f"M={len(branch_feats)} (served: C_out=1, G<=64, M<=3)")
     return ref.branch_fuse_head(branch_feats, fc_weight, fc_bias)
 
 

@@ -60,8 +60,9 @@ class ModelTrainer:
         # ragged last batches and eval run eagerly. Requires the FusedAdam
         # path (device-side bias correction) on GPU.
         self.use_graph = use_graph and torch.cuda.is_available()
-        self._graph = None            # (graph, x_static, y_static, loss_static)
-        self._graph_shape = None
+        self._graphs = {}             # shape -> (graph, x_static, y_static, loss_static)
+        self._eager_seen = set()      # shapes that ran their one eager warmup
+        self._resume_best_val = np.inf
 
     # ------------------------------------------------------------------ utils
     def _log(self, *args):
@@ -112,40 +113,42 @@ class ModelTrainer:
         """Returns the step loss as a FLOAT — never a live tensor: a loss
         held by the caller across iterations keeps the autograd graph (and
         its default-stream AccumulateGrad nodes) alive, which breaks hipGraph
-        capture of the next step. Replays the captured whole-step graph for
-        full-size batches; captures lazily on the second sighting of a shape
-        (after one eager warmup); anything else runs eagerly."""
+        capture of the next step. Graphs are cached PER SHAPE (the ragged
+        last batch of an epoch gets its own graph and never clobbers the
+        full-batch one); each shape pays one eager warmup step, then its
+        second sighting is the warmup-before-capture step — the capture
+        itself records without executing and is NOT replayed on that batch,
+        so every batch is trained exactly once."""
         if not self.use_graph:
             return float(self._train_step_eager(x, y_true, sta_adj_list).detach())
         shape = (tuple(x.shape), tuple(y_true.shape))
-        if self._graph is not None and self._graph_shape == shape:
-            g, xs, ys, ls = self._graph
+        entry = self._graphs.get(shape)
+        if entry is not None:
+            g, xs, ys, ls = entry
             xs.copy_(x)
             ys.copy_(y_true)
             g.replay()
             return float(ls.detach())
-        if self._graph_shape != shape:           # first sighting: eager warmup
-            self._graph_shape = shape
-            self._graph = None
+        if shape not in self._eager_seen:        # first sighting: eager warmup
+            self._eager_seen.add(shape)
             return float(self._train_step_eager(x, y_true, sta_adj_list).detach())
-        # second sighting of the same shape: capture
+        # second sighting of the shape: the side-stream warmup IS this
+        # batch's (single) real training step; then record the graph.
         try:
             xs, ys = x.clone(), y_true.clone()
             side = torch.cuda.Stream()
             side.wait_stream(torch.cuda.current_stream())
             with torch.cuda.stream(side):
-                self._train_step_eager(xs, ys, sta_adj_list)
+                warm_loss = self._train_step_eager(xs, ys, sta_adj_list)
             torch.cuda.current_stream().wait_stream(side)
             torch.cuda.synchronize()
+            warm = float(warm_loss.detach())
             g = torch.cuda.CUDAGraph()
             with torch.cuda.graph(g):
                 ls = self._train_step_eager(xs, ys, sta_adj_list)
             torch.cuda.synchronize()
-            self._graph = (g, xs, ys, ls)
-            xs.copy_(x)
-            ys.copy_(y_true)
-            g.replay()
-            return float(ls.detach())
+            self._graphs[shape] = (g, xs, ys, ls)
+            return warm
         except Exception as e:
             self._log(f"# hipGraph capture unavailable, eager fallback: {e}")
             self.use_graph = False
@@ -173,8 +176,14 @@ class ModelTrainer:
             {k: v.to(device=dev, dtype=dt) for k, v in saved["state_dict"].items()})
         opath = self._optim_path(model_dir)
         if os.path.exists(opath):
-            self.optimizer.load_state_dict(
-                torch.load(opath, weights_only=False, map_location=dev))
+            payload = torch.load(opath, weights_only=False, map_location=dev)
+            if isinstance(payload, dict) and "optimizer" in payload:
+                self.optimizer.load_state_dict(payload["optimizer"])
+                # seed the best-val floor so the first post-resume epoch
+                # cannot overwrite a better pre-crash checkpoint
+                self._resume_best_val = float(payload.get("best_val_loss", np.inf))
+            else:                     # legacy sidecar: bare optimizer state
+                self.optimizer.load_state_dict(payload)
         self._log(f"Resumed from {path} (epoch {saved['epoch']})")
         return int(saved["epoch"])
 
@@ -182,8 +191,12 @@ class ModelTrainer:
     def train(self, data_loader: Dict, sta_adj_list: List, modes: List[str],
               model_dir: str, early_stopper: int = 10, start_epoch: int = 0):
         patience = early_stopper
-        checkpoint = {"epoch": start_epoch, "state_dict": self.model.state_dict()}
-        val_loss = np.inf
+        # cloned snapshot (not live references): the unconditional final save
+        # must never write weights from a later epoch under this epoch number
+        checkpoint = {"epoch": start_epoch,
+                      "state_dict": {k: v.detach().clone()
+                                     for k, v in self.model.state_dict().items()}}
+        val_loss = self._resume_best_val
         self._log("Training starts at: ", time.ctime())
 
         for epoch in range(start_epoch + 1, self.n_epochs + 1):
@@ -216,10 +229,18 @@ class ModelTrainer:
                         self._log(f"Epoch {epoch}, Val_loss drops from {val_loss:.5} "
                                   f"to {epoch_val:.5}. Update model checkpoint..")
                         val_loss = epoch_val
-                        checkpoint.update(epoch=epoch, state_dict=self.model.state_dict())
+                        # snapshot (clone) the best-epoch weights: the
+                        # reference keeps live references here, so its final
+                        # save silently writes LAST-epoch weights under the
+                        # best epoch number — quirk fixed like 6/7
+                        checkpoint.update(
+                            epoch=epoch,
+                            state_dict={k: v.detach().clone()
+                                        for k, v in self.model.state_dict().items()})
                         if self.rank == 0:
                             torch.save(checkpoint, self._ckpt_path(model_dir))
-                            torch.save(self.optimizer.state_dict(),
+                            torch.save({"optimizer": self.optimizer.state_dict(),
+                                        "best_val_loss": val_loss},
                                        self._optim_path(model_dir))
                         patience = early_stopper
                     else:

@@ -43,3 +43,30 @@ def test_bench_json_contract():
 def test_bench_torch_floor_mode():
     rec = _run_bench("--impl", "torch")
     assert rec["config"]["impl"].startswith("torch")
+
+
+def test_bench_torchrun_world2_cpu():
+    """The driver's exact multi-rank launch (torch.distributed.run, nnodes=1,
+    one JSON line from rank 0) works end-to-end: rendezvous over 127.0.0.1,
+    gloo on CPU, max-over-ranks elapsed reduce, global-batch reporting.
+    Pre-hardens the 8-GPU SCALE run (VERDICT r1 next #6)."""
+    import socket
+    s = socket.socket()
+    s.bind(("127.0.0.1", 0))
+    port = s.getsockname()[1]
+    s.close()
+    r = subprocess.run(
+        [sys.executable, "-m", "torch.distributed.run", "--nnodes=1",
+         "--nproc-per-node", "2", "--master-addr", "127.0.0.1",
+         "--master-port", str(port), os.path.join(REPO, "bench.py"),
+         "--gpus", "2", "--steps", "2", "--warmup", "1",
+         "--preset", "cpu-small"],
+        capture_output=True, text=True, timeout=600, cwd=REPO)
+    assert r.returncode == 0, r.stderr[-2000:]
+    json_lines = [l for l in r.stdout.splitlines() if l.startswith("{")]
+    assert len(json_lines) == 1, f"expected exactly one JSON line: {r.stdout}"
+    rec = json.loads(json_lines[0])
+    assert rec["n_gpus"] == 2
+    assert rec["config"]["parallelism"] == "dp2"
+    assert rec["config"]["global_batch"] == 16      # 8 per rank x 2 ranks
+    assert rec["value"] > 0

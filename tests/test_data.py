@@ -111,3 +111,26 @@ def test_get_data_loader_short_synthetic_rescales():
     assert total == 24 * 30 - 168
     for mode in ["train", "validate", "test"]:
         assert len(loaders[mode]) > 0
+
+
+def test_device_loader_drop_last_when_sharded():
+    """Sharded (world>1) loaders DROP the ragged final global batch — a
+    partial batch cannot be split evenly across ranks and would desync the
+    per-rank step counts (the all-reduce would deadlock). Unsharded loaders
+    keep it (reference DataLoader semantics)."""
+    import torch
+    from stmgcn_amd.data.container import DeviceLoader
+    n, b, world = 21, 4, 2                 # global batch 8 -> 2 full, 5 dropped
+    x = torch.arange(n, dtype=torch.float32).reshape(n, 1)
+    y = x.clone()
+    counts = []
+    for rank in range(world):
+        dl = DeviceLoader(x, y, batch_size=b, rank=rank, world_size=world)
+        batches = list(dl)
+        counts.append(len(batches))
+        assert all(xb.shape[0] == b for xb, _ in batches)   # only full batches
+    assert counts[0] == counts[1] == len(dl) == n // (b * world)
+    # unsharded: ragged tail kept
+    dl1 = DeviceLoader(x, y, batch_size=b)
+    sizes = [xb.shape[0] for xb, _ in dl1]
+    assert sum(sizes) == n and sizes[-1] == n % b

@@ -164,3 +164,28 @@ def test_checkpoint_roundtrip(tmp_path):
     for (n1, p1), (n2, p2) in zip(model.named_parameters(), model2.named_parameters()):
         assert n1 == n2
         torch.testing.assert_close(p1, p2)
+
+
+def test_nn_module_activation_accepted():
+    """The reference wires gconv_activation=nn.ReLU (Main.py:64); the ctor
+    accepts the module form and it matches the string form exactly."""
+    from torch import nn
+    torch.manual_seed(0)
+    m1 = ST_MGCN(M=1, seq_len=4, n_nodes=12, input_dim=1, lstm_hidden_dim=8,
+                 lstm_num_layers=1, gcn_hidden_dim=8,
+                 sta_kernel_config={"kernel_type": "chebyshev", "K": 2},
+                 gconv_activation=nn.ReLU)
+    torch.manual_seed(0)
+    m2 = ST_MGCN(M=1, seq_len=4, n_nodes=12, input_dim=1, lstm_hidden_dim=8,
+                 lstm_num_layers=1, gcn_hidden_dim=8,
+                 sta_kernel_config={"kernel_type": "chebyshev", "K": 2},
+                 gconv_activation="relu")
+    gen = SupportGenerator("chebyshev", 2)
+    adjs = [gen.process(a) for a in _adjs(12, m=1)]
+    x = torch.randn(2, 4, 12, 1)
+    torch.testing.assert_close(m1(x, adjs), m2(x, adjs))
+    with pytest.raises(ValueError, match="activation"):
+        ST_MGCN(M=1, seq_len=4, n_nodes=12, input_dim=1, lstm_hidden_dim=8,
+                lstm_num_layers=1, gcn_hidden_dim=8,
+                sta_kernel_config={"kernel_type": "chebyshev", "K": 2},
+                gconv_activation=torch.nn.Tanh)

@@ -95,20 +95,6 @@ def test_trainer_rejects_unknown_model():
                      optimizer=optim.Adam, lr=1e-3, wd=0, n_epochs=1)
 
 
-def test_main_cli_end_to_end(tmp_path):
-    """Main.py runs a tiny synthetic training + test on CPU."""
-    out = subprocess.run(
-        [sys.executable, os.path.join(REPO, "Main.py"),
-         "-device", "cpu", "--synthetic", "--preset", "cpu-small",
-         "--nodes", "9", "--epochs", "1", "--batch-size", "16",
-         "--model-dir", str(tmp_path)],
-        capture_output=True, text=True, timeout=600, cwd=REPO)
-    assert out.returncode == 0, out.stderr[-2000:]
-    assert "Training starts at:" in out.stdout
-    assert "true RMSE:" in out.stdout
-    assert (tmp_path / "ST_MGCN_best_model.pkl").exists()
-
-
 def test_main_cli_sparse_path(tmp_path):
     out = subprocess.run(
         [sys.executable, os.path.join(REPO, "Main.py"),
@@ -210,9 +196,37 @@ def test_resume_from_best_checkpoint(tmp_path):
     for k, v in t2.model.state_dict().items():
         torch.testing.assert_close(v, saved["state_dict"][k])
     assert t2.optimizer.state_dict()["state"]          # optimizer moments loaded
+    # the sidecar seeds the best-val floor: the first post-resume epoch must
+    # not overwrite a better pre-crash checkpoint with a worse one
+    assert np.isfinite(t2._resume_best_val)
     t2.n_epochs = start + 1
     t2.train(loaders, adjs, ["train", "validate"], str(tmp_path),
              start_epoch=start)                        # continues, no crash
+
+
+def test_resume_does_not_clobber_better_checkpoint(tmp_path):
+    """Crash/resume cycle: if every post-resume epoch is WORSE than the
+    pre-crash best, the best checkpoint file must survive untouched
+    (train() used to restart at val_loss=inf and overwrite)."""
+    make, loaders, adjs = _tiny_setup(seed=5)
+    t1 = make(n_epochs=2, lr=1e-2)
+    vals1 = iter([1.0, 0.25])             # best = epoch 2 @ 0.25
+    t1._allreduce_scalar = lambda v: float(next(vals1))
+    t1.train(loaders, adjs, ["train", "validate"], str(tmp_path))
+    best = torch.load(tmp_path / "ST_MGCN_best_model.pkl", weights_only=False)
+    assert best["epoch"] == 2
+
+    t2 = make(n_epochs=4, lr=1e-2)        # "restarted process"
+    start = t2.resume(str(tmp_path))
+    assert t2._resume_best_val == pytest.approx(0.25)
+    vals2 = iter([0.9, 1.5])              # both worse than 0.25
+    t2._allreduce_scalar = lambda v: float(next(vals2))
+    t2.train(loaders, adjs, ["train", "validate"], str(tmp_path),
+             start_epoch=start)
+    after = torch.load(tmp_path / "ST_MGCN_best_model.pkl", weights_only=False)
+    assert after["epoch"] == 2            # pre-crash best not overwritten
+    for k, v in after["state_dict"].items():
+        torch.testing.assert_close(v, best["state_dict"][k])
 
 
 def _tiny_setup(seed=0):
@@ -261,17 +275,20 @@ def test_early_stopping_fires(tmp_path, capsys):
 
 
 def test_final_save_is_best_not_last(tmp_path):
-    """The end-of-training save re-saves the BEST checkpoint dict, not the
-    last epoch's weights (reference quirk 10 semantics)."""
+    """The end-of-training save preserves the BEST-epoch weights even when a
+    later, worse epoch ran afterwards. (The reference aliases live tensors in
+    its checkpoint dict, so its final save silently writes last-epoch weights
+    under the best epoch number — fixed by cloning at best-epoch time.)
+    Forced val losses make best=epoch 2, last=epoch 3 deterministically."""
     make, loaders, adjs = _tiny_setup(seed=3)
-    tr = make(n_epochs=3)
+    tr = make(n_epochs=3, lr=1e-2)
+    vals = iter([1.0, 0.5, 2.0])          # improves at 2, worsens at 3
+    tr._allreduce_scalar = lambda v: float(next(vals))
     tr.train(loaders, adjs, ["train", "validate"], str(tmp_path),
              early_stopper=50)
     saved = torch.load(tmp_path / "ST_MGCN_best_model.pkl", weights_only=False)
-    # the saved epoch is the best-val epoch; weights differ from the live
-    # (last-epoch) model unless the best WAS the last epoch
-    assert 1 <= saved["epoch"] <= 3
-    if saved["epoch"] != 3:
-        live = tr.model.state_dict()
-        assert any(not torch.equal(v, live[k])
-                   for k, v in saved["state_dict"].items())
+    assert saved["epoch"] == 2
+    live = tr.model.state_dict()
+    # epoch 3 trained with lr>0 -> live weights moved past the saved snapshot
+    assert any(not torch.equal(v, live[k])
+               for k, v in saved["state_dict"].items())
