@@ -133,6 +133,173 @@ at::Tensor cheb_combine(at::Tensor U, at::Tensor rowptr, at::Tensor colidx,
 }
 
 // ---------------------------------------------------------------------------
+// Fully fused ChebConv (SURVEY K1+K2): recurrence + MFMA mix + bias + act in
+// one kernel chain; no support stack, no library GEMMs.
+
+extern "C" void stmgcn_cheb_fused_fwd_step(
+    void* stream, int dtype, const int* rowptr, const int* colidx,
+    const float* vals, const void* xin, const void* p1, const void* W,
+    const void* bias, void* pout, float* yacc, void* yout, int B, int N,
+    int C, int Cout, int kofs, float alpha, float beta, int first, int act);
+extern "C" void stmgcn_cheb_fused_bwd_step(
+    void* stream, int dtype, const int* rowptr, const int* colidx,
+    const float* vals, const void* xin, const void* p1, const void* dz,
+    const void* W, void* out, int B, int N, int C, int Cout, int kofs,
+    float alpha, float beta);
+
+// y = act(sum_k (T_k(G) x) @ W_k + b), fused: the K_s recurrence steps each
+// fold their mix into an fp32 accumulator; the last adds bias + activation.
+// Serves C <= 64, Cout <= 64 (every BASELINE config; the stack path in
+// cheb_apply remains for parity tests / larger widths).
+at::Tensor cheb_gconv_fused_fwd(at::Tensor x, at::Tensor rowptr,
+                                at::Tensor colidx, at::Tensor vals,
+                                at::Tensor W, c10::optional<at::Tensor> bias,
+                                int64_t K_s, bool single, int64_t act) {
+  TORCH_CHECK(x.is_cuda() && x.dim() == 3 && x.is_contiguous());
+  TORCH_CHECK(W.is_contiguous() && W.scalar_type() == x.scalar_type());
+  const int B = x.size(0), N = x.size(1), C = x.size(2);
+  const int Cout = W.size(1);
+  TORCH_CHECK(C <= 64 && Cout <= 64, "fused ChebConv serves C/Cout <= 64");
+  TORCH_CHECK(W.size(0) == K_s * C);
+  const int dt = dtype_code(x);
+  void* st = stream();
+  const int* rp = rowptr.data_ptr<int>();
+  const int* ci = colidx.data_ptr<int>();
+  const float* v = vals.data_ptr<float>();
+  auto y = at::empty({B, N, Cout}, x.options());
+  const void* bp = bias.has_value() ? bias->contiguous().data_ptr() : nullptr;
+  const int actc = (int)act;
+
+  if (single) {  // localpool: y = act((G @ x) @ W + b)
+    TORCH_CHECK(K_s == 1);
+    stmgcn_cheb_fused_fwd_step(st, dt, rp, ci, v, x.data_ptr(), nullptr,
+                               W.data_ptr(), bp, nullptr, nullptr,
+                               y.data_ptr(), B, N, C, Cout, 0, 1.f, 0.f, 1,
+                               actc);
+    return y;
+  }
+  if (K_s == 1) {  // T_0 only: y = act(x @ W + b)
+    stmgcn_cheb_fused_fwd_step(st, dt, rp, ci, v, nullptr, x.data_ptr(),
+                               W.data_ptr(), bp, nullptr, nullptr,
+                               y.data_ptr(), B, N, C, Cout, 0, 0.f, 1.f, 1,
+                               actc);
+    return y;
+  }
+  auto yacc = at::empty({B, N, Cout}, x.options().dtype(at::kFloat));
+  float* ya = yacc.data_ptr<float>();
+  // k = 0: p_0 aliases x; mix only
+  stmgcn_cheb_fused_fwd_step(st, dt, rp, ci, v, nullptr, x.data_ptr(),
+                             W.data_ptr(), bp, nullptr, ya, nullptr, B, N, C,
+                             Cout, 0, 0.f, 1.f, 1, actc);
+  // k = 1: p_1 = G x
+  auto pA = at::empty({B, N, C}, x.options());
+  at::Tensor pB;
+  bool last = (K_s == 2);
+  stmgcn_cheb_fused_fwd_step(st, dt, rp, ci, v, x.data_ptr(), nullptr,
+                             W.data_ptr(), bp, pA.data_ptr(), ya,
+                             last ? y.data_ptr() : nullptr, B, N, C, Cout,
+                             C, 1.f, 0.f, 0, actc);
+  // k >= 2: p_k = 2 G p_{k-1} - p_{k-2}; p1 == pout aliasing is element-safe
+  const void* pm2 = x.data_ptr();
+  void* pm1 = pA.data_ptr();
+  for (int k = 2; k < K_s; ++k) {
+    void* dst;
+    if (k == 2) {
+      pB = at::empty({B, N, C}, x.options());
+      dst = pB.data_ptr();
+    } else {
+      dst = const_cast<void*>(pm2);  // overwrite p_{k-2} (element-wise read)
+    }
+    last = (k == K_s - 1);
+    stmgcn_cheb_fused_fwd_step(st, dt, rp, ci, v, pm1, pm2, W.data_ptr(), bp,
+                               dst, ya, last ? y.data_ptr() : nullptr, B, N,
+                               C, Cout, k * C, 2.f, -1.f, 0, actc);
+    pm2 = pm1;
+    pm1 = dst;
+  }
+  return y;
+}
+
+// dX = sum_k T_k(G)^T (dz W_k^T) via Clenshaw over the G^T CSR with the
+// U_j = dz W_j^T term fused into every step (never materialized).
+at::Tensor cheb_gconv_fused_bwd_dx(at::Tensor dz, at::Tensor W,
+                                   at::Tensor rowptr_t, at::Tensor colidx_t,
+                                   at::Tensor vals_t, int64_t K_s,
+                                   bool single) {
+  TORCH_CHECK(dz.is_cuda() && dz.dim() == 3 && dz.is_contiguous());
+  TORCH_CHECK(W.is_contiguous() && W.scalar_type() == dz.scalar_type());
+  const int B = dz.size(0), N = dz.size(1), Cout = dz.size(2);
+  const int C = W.size(0) / K_s;
+  TORCH_CHECK(C <= 64 && Cout <= 64 && (int)(K_s * C) == W.size(0));
+  const int dt = dtype_code(dz);
+  void* st = stream();
+  const int* rp = rowptr_t.data_ptr<int>();
+  const int* ci = colidx_t.data_ptr<int>();
+  const float* v = vals_t.data_ptr<float>();
+  auto dX = at::empty({B, N, C}, dz.options());
+
+  auto bwd = [&](const void* xin, const void* p1, void* out, int kofs,
+                 float alpha, float beta) {
+    stmgcn_cheb_fused_bwd_step(st, dt, rp, ci, v, xin, p1, dz.data_ptr(),
+                               W.data_ptr(), out, B, N, C, Cout, kofs, alpha,
+                               beta);
+  };
+  if (single) {  // dX = G^T U_0: materialize U_0, then one plain SpMM
+    auto U0 = at::empty({B, N, C}, dz.options());
+    bwd(nullptr, nullptr, U0.data_ptr(), 0, 0.f, 0.f);
+    stmgcn_cheb_fused_fwd_step(st, dt, rp, ci, v, U0.data_ptr(), nullptr,
+                               nullptr, nullptr, dX.data_ptr(), nullptr,
+                               nullptr, B, N, C, C, 0, 1.f, 0.f, 1, 0);
+    return dX;
+  }
+  const int K = (int)K_s - 1;
+  if (K == 0) {  // dX = U_0
+    bwd(nullptr, nullptr, dX.data_ptr(), 0, 0.f, 0.f);
+    return dX;
+  }
+  if (K == 1) {  // dX = U_0 + G^T U_1
+    auto b1 = at::empty({B, N, C}, dz.options());
+    bwd(nullptr, nullptr, b1.data_ptr(), C, 0.f, 0.f);
+    bwd(b1.data_ptr(), nullptr, dX.data_ptr(), 0, 1.f, 0.f);
+    return dX;
+  }
+  // general Clenshaw: b_K = U_K; b_j = U_j + 2 G^T b_{j+1} - b_{j+2};
+  // dX = U_0 + G^T b_1 - b_2. Two ping/pong buffers; p1 aliasing is safe.
+  auto bufA = at::empty({B, N, C}, dz.options());
+  auto bufB = at::empty({B, N, C}, dz.options());
+  void* bj1 = bufA.data_ptr();  // b_{j+1}
+  void* bj2 = nullptr;          // b_{j+2}
+  bwd(nullptr, nullptr, bj1, K * C, 0.f, 0.f);  // b_K
+  void* other = bufB.data_ptr();
+  for (int j = K - 1; j >= 1; --j) {
+    bwd(bj1, bj2, other, j * C, 2.f, bj2 ? -1.f : 0.f);
+    bj2 = bj1;
+    bj1 = other;
+    other = bj2;  // overwritten next iteration (element-wise p1 read is safe)
+  }
+  bwd(bj1, bj2, dX.data_ptr(), 0, 1.f, -1.f);
+  return dX;
+}
+
+// Plain axpby-SpMM on (B,N,C): out = alpha*(G @ xin) + beta*p1 — the
+// recurrence-replay primitive for the fused wgrad (backward recompute).
+at::Tensor spmm_axpby(at::Tensor xin, c10::optional<at::Tensor> p1,
+                      at::Tensor rowptr, at::Tensor colidx, at::Tensor vals,
+                      double alpha, double beta) {
+  TORCH_CHECK(xin.is_cuda() && xin.dim() == 3 && xin.is_contiguous());
+  const int B = xin.size(0), N = xin.size(1), C = xin.size(2);
+  auto out = at::empty_like(xin);
+  stmgcn_cheb_fused_fwd_step(stream(), dtype_code(xin),
+                             rowptr.data_ptr<int>(), colidx.data_ptr<int>(),
+                             vals.data_ptr<float>(), xin.data_ptr(),
+                             p1.has_value() ? p1->data_ptr() : nullptr,
+                             nullptr, nullptr, out.data_ptr(), nullptr,
+                             nullptr, B, N, C, C, 0, (float)alpha,
+                             (float)beta, 1, 0);
+  return out;
+}
+
+// ---------------------------------------------------------------------------
 // Fused LSTM: forward kernel + dgrad kernel; weight grads
 // are computed on the python side as plain library GEMMs over the dA stream.
 
@@ -268,6 +435,28 @@ std::vector<at::Tensor> atb_wgrad(at::Tensor A, at::Tensor B, bool want_db) {
                    C.data_ptr<float>(), want_db ? db.data_ptr<float>() : nullptr,
                    rows, M, N);
   return want_db ? std::vector<at::Tensor>{C, db} : std::vector<at::Tensor>{C};
+}
+
+// Accumulating variant: C (M,N) fp32 += A^T @ B, db (N,) fp32 += colsum(B).
+// Caller provides (zeroed or accumulating) outputs — used by the fused
+// ChebConv wgrad to land each support's dW_k slice in one flat buffer.
+void atb_wgrad_into(at::Tensor A, at::Tensor B, at::Tensor C,
+                    c10::optional<at::Tensor> db) {
+  TORCH_CHECK(A.is_cuda() && A.dim() == 2 && B.dim() == 2);
+  TORCH_CHECK(A.is_contiguous() && B.is_contiguous() && C.is_contiguous());
+  TORCH_CHECK(C.scalar_type() == at::kFloat);
+  const long rows = A.size(0);
+  const int M = A.size(1), N = B.size(1);
+  TORCH_CHECK(B.size(0) == rows && M <= 256 && N <= 64 && M % 8 == 0 && N % 8 == 0);
+  TORCH_CHECK(C.size(0) == M && C.size(1) == N);
+  float* dbp = nullptr;
+  if (db.has_value()) {
+    TORCH_CHECK(db->is_contiguous() && db->scalar_type() == at::kFloat &&
+                db->numel() == N);
+    dbp = db->data_ptr<float>();
+  }
+  stmgcn_atb_wgrad(stream(), dtype_code(A), A.data_ptr(), B.data_ptr(),
+                   C.data_ptr<float>(), dbp, rows, M, N);
 }
 
 at::Tensor mfma_probe(at::Tensor A, at::Tensor B) {
@@ -457,4 +646,12 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
         "Support stack S[b,n,k,c] = (T_k(G) x)[b,n,c] via in-kernel recurrence");
   m.def("cheb_combine", &cheb_combine,
         "Z = sum_k T_k(G) U_k via Clenshaw (pass G^T CSR for gradients)");
+  m.def("cheb_gconv_fused_fwd", &cheb_gconv_fused_fwd,
+        "Fused ChebConv fwd: recurrence + MFMA mix + bias + act, no stack");
+  m.def("cheb_gconv_fused_bwd_dx", &cheb_gconv_fused_bwd_dx,
+        "Fused ChebConv dX: Clenshaw over G^T with in-kernel U = dz W^T");
+  m.def("spmm_axpby", &spmm_axpby,
+        "out = alpha*(G @ xin) + beta*p1 on (B,N,C) (recurrence replay)");
+  m.def("atb_wgrad_into", &atb_wgrad_into,
+        "C += A^T B (+db += colsum B) into caller-provided fp32 buffers");
 }

@@ -36,6 +36,15 @@ __device__ __forceinline__ float stm_tanh(float x) {
   return fmaf(2.0f, stm_sigmoid(2.0f * t), -1.0f);
 }
 
+// LDS swizzle for 128-byte tile rows (64 bf16/f16 channels): element (row s,
+// channel byte cbyte) of one tile slot. XOR'd so the 16 lanes of a
+// ds_read_b128 fragment group (consecutive rows, same channel window) hit 64
+// distinct banks: row parity gives the 32-bank half, ((s>>1)&7) permutes the
+// eight 16 B windows within it.
+__device__ __forceinline__ int lds_swz(int s, int cbyte) {
+  return s * 128 + (cbyte ^ ((((unsigned)s >> 1) & 7) << 4));
+}
+
 #define STM_CHECK_HIP(expr)                                                     \
   do {                                                                          \
     hipError_t _e = (expr);                                                     \

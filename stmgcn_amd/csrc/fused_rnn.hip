@@ -44,16 +44,8 @@ struct RnnPtrs {
   const void* b_hh[MAX_LAYERS];   // (G*H,) model dtype
 };
 
-// ---------------------------------------------------------------------------
-// LDS swizzle: element (row s, channel c) of one timestep slot (128 B/row).
-// Byte = s*128 + c*2, XOR'd so the 16 lanes of a ds_read_b128 group
-// (consecutive rows, same channel window) hit 64 distinct banks: row parity
-// gives the 32-bank half, ((s>>1)&7) permutes the eight 16 B windows within
-// it — rows s and s+8 land in different windows (the previous (s&7) swizzle
-// collided them 2-way).
-__device__ __forceinline__ int lds_swz(int s, int cbyte) {
-  return s * 128 + (cbyte ^ ((((unsigned)s >> 1) & 7) << 4));
-}
+// (LDS swizzle lds_swz for 128 B tile rows lives in common.h — shared with
+// the fused ChebConv kernels.)
 
 template <typename T> struct Frag8;
 template <> struct Frag8<__hip_bfloat16> { using type = bf16x8; using elem = __bf16; };

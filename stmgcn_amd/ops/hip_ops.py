@@ -54,13 +54,20 @@ def _fallback(what: str) -> None:
 
 
 class ChebGconvFn(torch.autograd.Function):
-    """y = act(concat_k(T_k(G) x) @ W + b) with the K-hop Chebyshev
-    recurrence computed by the HIP kernel on CSR G (never materializing T_k;
-    reference materializes dense stacks, GCN.py:95/34-42). Mix GEMMs go to
-    rocBLAS (plain library GEMMs); dW/db likewise.
+    """y = act(concat_k(T_k(G) x) @ W + b) — fully fused ChebConv
+    (reference GCN.py:34-42 materializes dense (K,N,N)
+    stacks offline and a (B,N,K_s,C) feature concat per forward).
 
-    Backward: dX = sum_k T_k(G)^T (dZ W_k^T) = cheb_combine over G^T CSR
-    (Clenshaw); dW = feat^T dZ; db = sum dZ.
+    Fused path (bf16/f16, C/Cout <= 64 and mult-of-8 — every BASELINE
+    config): the K_s recurrence steps run in-kernel with the mix GEMM +
+    bias + ReLU folded into each step's MFMA epilogue; nothing but x and y
+    is saved. Backward: dX via Clenshaw over G^T with U_j = dz W_j^T fused
+    in-kernel; dW_k = (T_k x)^T dz by REPLAYING the recurrence (spmm_axpby)
+    and feeding each state to the atb_wgrad MFMA reduction kernel. Zero
+    library GEMMs, zero (B,N,K_s,C) stacks in either direction.
+
+    Stack path (fp32 parity / larger widths): cheb_apply support stack +
+    rocBLAS mix, Clenshaw combine backward.
     """
 
     @staticmethod
@@ -68,13 +75,31 @@ class ChebGconvFn(torch.autograd.Function):
                 csr: CSRSupport, activation: Optional[str]):
         C = require_hip()
         x = x.contiguous()
+        Cin = x.shape[-1]
+        Cout = W.shape[1]
+        Wd = W.to(x.dtype).contiguous()
+        fused = (x.dtype in (torch.bfloat16, torch.float16)
+                 and Cin <= 64 and Cout <= 64
+                 and Cin % 8 == 0 and Cout % 8 == 0)
+        ctx.fused = fused
+        ctx.csr = csr
+        ctx.act = activation
+        ctx.has_b = b is not None
+        ctx.cin = Cin
+        if fused:
+            y = C.cheb_gconv_fused_fwd(
+                x, csr.row_ptr, csr.col_idx, csr.vals, Wd,
+                b.to(x.dtype).contiguous() if b is not None else None,
+                csr.K_supports, csr.kind == "single",
+                1 if activation == "relu" else 0)
+            ctx.save_for_backward(x, Wd, W,
+                                  y if activation == "relu" else None)
+            return y
         S = C.cheb_apply(x, csr.row_ptr, csr.col_idx, csr.vals,
                          csr.K_supports, csr.kind == "single")
-        B_, N, K, Cin = S.shape
+        B_, N, K, _ = S.shape
         feat = S.view(B_, N, K * Cin)
-        Wd = W.to(feat.dtype)
         if b is not None:
-            # bias rides the hipBLASLt GEMM epilogue (one launch, no add)
             y = torch.addmm(b.to(feat.dtype), feat.view(-1, K * Cin), Wd)
             y = y.view(B_, N, -1)
         else:
@@ -82,17 +107,53 @@ class ChebGconvFn(torch.autograd.Function):
         if activation == "relu":
             y = torch.relu_(y)
         ctx.save_for_backward(feat, W, y if activation == "relu" else None)
-        ctx.csr = csr
-        ctx.act = activation
-        ctx.has_b = b is not None
-        ctx.cin = Cin
         return y
 
     @staticmethod
     def backward(ctx, dy: torch.Tensor):
         C = require_hip()
-        feat, W, y = ctx.saved_tensors
         csr: CSRSupport = ctx.csr
+        if ctx.fused:
+            x, Wd, W, y = ctx.saved_tensors
+            if ctx.act == "relu":
+                dz = (dy * (y > 0).to(dy.dtype)).contiguous()
+            else:
+                dz = dy.contiguous()
+            single = csr.kind == "single"
+            K_s = csr.K_supports
+            Cin, Cout = ctx.cin, dz.shape[-1]
+            dX = C.cheb_gconv_fused_bwd_dx(dz, Wd, csr.row_ptr_t,
+                                           csr.col_idx_t, csr.vals_t, K_s,
+                                           single)
+            # ---- dW/db: replay the recurrence, one atb MFMA call per support
+            dWf = torch.zeros(K_s * Cin, Cout, dtype=torch.float32,
+                              device=dz.device)
+            db_f = (torch.zeros(Cout, dtype=torch.float32, device=dz.device)
+                    if ctx.has_b else None)
+            dz2 = dz.reshape(-1, Cout)
+            if single:
+                p = C.spmm_axpby(x, None, csr.row_ptr, csr.col_idx, csr.vals,
+                                 1.0, 0.0)
+                C.atb_wgrad_into(p.reshape(-1, Cin), dz2, dWf, db_f)
+            else:
+                C.atb_wgrad_into(x.reshape(-1, Cin), dz2, dWf[:Cin], db_f)
+                if K_s > 1:
+                    p1 = C.spmm_axpby(x, None, csr.row_ptr, csr.col_idx,
+                                      csr.vals, 1.0, 0.0)
+                    C.atb_wgrad_into(p1.reshape(-1, Cin), dz2,
+                                     dWf[Cin:2 * Cin], None)
+                    pm2, pm1 = x, p1
+                    for k in range(2, K_s):
+                        pk = C.spmm_axpby(pm1, pm2, csr.row_ptr, csr.col_idx,
+                                          csr.vals, 2.0, -1.0)
+                        C.atb_wgrad_into(pk.reshape(-1, Cin), dz2,
+                                         dWf[k * Cin:(k + 1) * Cin], None)
+                        pm2, pm1 = pm1, pk
+            dW = dWf.to(W.dtype)
+            db = db_f.to(W.dtype) if ctx.has_b else None
+            return dX, dW, db, None, None
+
+        feat, W, y = ctx.saved_tensors
         if ctx.act == "relu":
             dz = dy * (y > 0).to(dy.dtype)
         else:

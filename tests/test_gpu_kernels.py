@@ -636,3 +636,67 @@ def test_trainer_hipgraph_capture_path():
     # both runs converge; allow the extra capture-warmup steps' perturbation
     assert losses[True][-1] < 0.7 * losses[True][0]
     assert losses[False][-1] < 0.7 * losses[False][0]
+
+
+@pytest.mark.parametrize("dtype", [torch.bfloat16, torch.float16])
+@pytest.mark.parametrize("cin,cout,kernel,K", [
+    (64, 64, "chebyshev", 2),   # MFMA fwd epilogue + MFMA bwd U-GEMM
+    (8, 8, "chebyshev", 2),     # temporal-gconv shape (scalar epilogues)
+    (64, 64, "chebyshev", 3),   # K_s=4: ping/pong recurrence + deep Clenshaw
+    (16, 64, "localpool", 2),   # single-support path
+    (64, 32, "chebyshev", 2),   # MFMA fwd, scalar bwd (Cout % 32 != 0)
+])
+def test_cheb_gconv_fused_parity(dtype, cin, cout, kernel, K):
+    """Fully fused ChebConv (no support stack, no library GEMMs) vs the fp32
+    CPU oracle: forward, dX, dW, db (VERDICT r1 next #2)."""
+    from stmgcn_amd.ops.hip_ops import ChebGconvFn
+    csr = _csr(kernel=kernel, K=K)
+    dev = torch.device("cuda")
+    csr_d = csr.to(dev)
+    B, N = 3, csr.n_nodes
+    torch.manual_seed(1)
+    x = torch.randn(B, N, cin, device=dev, dtype=dtype, requires_grad=True)
+    W = (torch.randn(csr.K_supports * cin, cout, device=dev, dtype=dtype,
+                     requires_grad=True) * 0.15).detach().requires_grad_(True)
+    b = torch.randn(cout, device=dev, dtype=dtype) * 0.1
+    b.requires_grad_(True)
+    y = ChebGconvFn.apply(x, W, b, csr_d, "relu")
+    (y.float() ** 2).sum().backward()
+
+    x_ref = x.detach().float().cpu().requires_grad_(True)
+    W_ref = W.detach().float().cpu().requires_grad_(True)
+    b_ref = b.detach().float().cpu().requires_grad_(True)
+    y_ref = ref.gconv_mix_csr(csr, x_ref, W_ref, b_ref, "relu")
+    (y_ref ** 2).sum().backward()
+
+    def relerr(a, r):
+        return ((a.float().cpu() - r).abs().max() / (r.abs().max() + 1e-6)).item()
+
+    assert relerr(y.detach(), y_ref.detach()) < 0.05, "fwd"
+    assert relerr(x.grad, x_ref.grad) < 0.08, "dX"
+    assert relerr(W.grad, W_ref.grad) < 0.08, "dW"
+    assert relerr(b.grad, b_ref.grad) < 0.08, "db"
+
+
+def test_cheb_gconv_fused_no_bias_no_act():
+    """Fused path with bias=None / activation=None (GCN(bias=False))."""
+    from stmgcn_amd.ops.hip_ops import ChebGconvFn
+    csr = _csr()
+    dev = torch.device("cuda")
+    csr_d = csr.to(dev)
+    torch.manual_seed(2)
+    x = torch.randn(2, csr.n_nodes, 64, device=dev, dtype=torch.bfloat16,
+                    requires_grad=True)
+    W = (torch.randn(csr.K_supports * 64, 64, device=dev,
+                     dtype=torch.bfloat16) * 0.15).requires_grad_(True)
+    y = ChebGconvFn.apply(x, W, None, csr_d, None)
+    (y.float() ** 2).sum().backward()
+    x_ref = x.detach().float().cpu().requires_grad_(True)
+    W_ref = W.detach().float().cpu().requires_grad_(True)
+    y_ref = ref.gconv_mix_csr(csr, x_ref, W_ref, None, None)
+    (y_ref ** 2).sum().backward()
+    for got, want in [(y.detach(), y_ref.detach()), (x.grad, x_ref.grad),
+                      (W.grad, W_ref.grad)]:
+        rel = ((got.float().cpu() - want).abs().max() /
+               (want.abs().max() + 1e-6)).item()
+        assert rel < 0.08
