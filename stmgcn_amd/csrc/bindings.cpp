@@ -151,10 +151,16 @@ extern "C" void stmgcn_cheb_fused_bwd_step(
 // fold their mix into an fp32 accumulator; the last adds bias + activation.
 // Serves C <= 64, Cout <= 64 (every BASELINE config; the stack path in
 // cheb_apply remains for parity tests / larger widths).
-at::Tensor cheb_gconv_fused_fwd(at::Tensor x, at::Tensor rowptr,
-                                at::Tensor colidx, at::Tensor vals,
-                                at::Tensor W, c10::optional<at::Tensor> bias,
-                                int64_t K_s, bool single, int64_t act) {
+//
+// training=true additionally returns the recurrence states p_1..p_{K_s-1}
+// (p_0 == x) for the backward wgrad: the ping/pong state writes are
+// mandatory anyway (cross-workgroup recurrence dependency), so keeping them
+// costs ZERO extra HBM traffic — unlike the reference-style (B,N,K_s,C)
+// concat stack, which existed only to feed a library GEMM and is gone.
+std::vector<at::Tensor> cheb_gconv_fused_fwd(
+    at::Tensor x, at::Tensor rowptr, at::Tensor colidx, at::Tensor vals,
+    at::Tensor W, c10::optional<at::Tensor> bias, int64_t K_s, bool single,
+    int64_t act, bool training) {
   TORCH_CHECK(x.is_cuda() && x.dim() == 3 && x.is_contiguous());
   TORCH_CHECK(W.is_contiguous() && W.scalar_type() == x.scalar_type());
   const int B = x.size(0), N = x.size(1), C = x.size(2);
@@ -170,20 +176,26 @@ at::Tensor cheb_gconv_fused_fwd(at::Tensor x, at::Tensor rowptr,
   const void* bp = bias.has_value() ? bias->contiguous().data_ptr() : nullptr;
   const int actc = (int)act;
 
-  if (single) {  // localpool: y = act((G @ x) @ W + b)
+  if (single) {  // localpool: y = act((G @ x) @ W + b); p = G x kept if training
     TORCH_CHECK(K_s == 1);
+    at::Tensor p;
+    void* pp = nullptr;
+    if (training) {
+      p = at::empty({B, N, C}, x.options());
+      pp = p.data_ptr();
+    }
     stmgcn_cheb_fused_fwd_step(st, dt, rp, ci, v, x.data_ptr(), nullptr,
-                               W.data_ptr(), bp, nullptr, nullptr,
-                               y.data_ptr(), B, N, C, Cout, 0, 1.f, 0.f, 1,
-                               actc);
-    return y;
+                               W.data_ptr(), bp, pp, nullptr, y.data_ptr(),
+                               B, N, C, Cout, 0, 1.f, 0.f, 1, actc);
+    return training ? std::vector<at::Tensor>{y, p}
+                    : std::vector<at::Tensor>{y};
   }
   if (K_s == 1) {  // T_0 only: y = act(x @ W + b)
     stmgcn_cheb_fused_fwd_step(st, dt, rp, ci, v, nullptr, x.data_ptr(),
                                W.data_ptr(), bp, nullptr, nullptr,
                                y.data_ptr(), B, N, C, Cout, 0, 0.f, 1.f, 1,
                                actc);
-    return y;
+    return {y};
   }
   auto yacc = at::empty({B, N, Cout}, x.options().dtype(at::kFloat));
   float* ya = yacc.data_ptr<float>();
@@ -191,25 +203,25 @@ at::Tensor cheb_gconv_fused_fwd(at::Tensor x, at::Tensor rowptr,
   stmgcn_cheb_fused_fwd_step(st, dt, rp, ci, v, nullptr, x.data_ptr(),
                              W.data_ptr(), bp, nullptr, ya, nullptr, B, N, C,
                              Cout, 0, 0.f, 1.f, 1, actc);
+  // k >= 1 recurrence states; training keeps every p_k (zero extra traffic —
+  // the writes are mandatory), eval ping/pongs two buffers.
+  std::vector<at::Tensor> ps;
+  const int nbuf = training ? (int)K_s - 1 : std::min<int>(2, (int)K_s - 1);
+  for (int i = 0; i < nbuf; ++i)
+    ps.push_back(at::empty({B, N, C}, x.options()));
   // k = 1: p_1 = G x
-  auto pA = at::empty({B, N, C}, x.options());
-  at::Tensor pB;
   bool last = (K_s == 2);
   stmgcn_cheb_fused_fwd_step(st, dt, rp, ci, v, x.data_ptr(), nullptr,
-                             W.data_ptr(), bp, pA.data_ptr(), ya,
+                             W.data_ptr(), bp, ps[0].data_ptr(), ya,
                              last ? y.data_ptr() : nullptr, B, N, C, Cout,
                              C, 1.f, 0.f, 0, actc);
   // k >= 2: p_k = 2 G p_{k-1} - p_{k-2}; p1 == pout aliasing is element-safe
   const void* pm2 = x.data_ptr();
-  void* pm1 = pA.data_ptr();
+  void* pm1 = ps[0].data_ptr();
   for (int k = 2; k < K_s; ++k) {
-    void* dst;
-    if (k == 2) {
-      pB = at::empty({B, N, C}, x.options());
-      dst = pB.data_ptr();
-    } else {
-      dst = const_cast<void*>(pm2);  // overwrite p_{k-2} (element-wise read)
-    }
+    void* dst = training ? ps[k - 1].data_ptr()
+                         : (k == 2 ? ps[1].data_ptr()
+                                   : const_cast<void*>(pm2));
     last = (k == K_s - 1);
     stmgcn_cheb_fused_fwd_step(st, dt, rp, ci, v, pm1, pm2, W.data_ptr(), bp,
                                dst, ya, last ? y.data_ptr() : nullptr, B, N,
@@ -217,7 +229,10 @@ at::Tensor cheb_gconv_fused_fwd(at::Tensor x, at::Tensor rowptr,
     pm2 = pm1;
     pm1 = dst;
   }
-  return y;
+  if (!training) return {y};
+  std::vector<at::Tensor> out{y};
+  for (auto& p : ps) out.push_back(p);
+  return out;
 }
 
 // dX = sum_k T_k(G)^T (dz W_k^T) via Clenshaw over the G^T CSR with the
@@ -324,6 +339,10 @@ extern "C" void stmgcn_lstm_wgrad(void* stream, int dtype, const void* dA,
 extern "C" void stmgcn_atb_wgrad(void* stream, int dtype, const void* A,
                                  const void* B, float* C, float* db, long rows,
                                  int M, int N);
+extern "C" void stmgcn_atb_wgrad_multi(void* stream, int dtype,
+                                       const void** As, int nsrc, int CA,
+                                       const void* B, float* C, float* db,
+                                       long rows, int N);
 
 static constexpr int kSeqTile = 32;  // MUST match SEQ_TILE in fused_rnn.hip
 static constexpr int kH = 64;
@@ -457,6 +476,36 @@ void atb_wgrad_into(at::Tensor A, at::Tensor B, at::Tensor C,
   }
   stmgcn_atb_wgrad(stream(), dtype_code(A), A.data_ptr(), B.data_ptr(),
                    C.data_ptr<float>(), dbp, rows, M, N);
+}
+
+// Multi-source reduction GEMM: C[k*CA:(k+1)*CA, :] += As[k]^T @ B for up to
+// 4 sources in ONE launch (B/dZ streamed once) + db += colsum(B). The fused
+// ChebConv wgrad feeds the recurrence states [x, p_1, ..] here.
+void atb_wgrad_multi(std::vector<at::Tensor> As, at::Tensor B, at::Tensor C,
+                     c10::optional<at::Tensor> db) {
+  TORCH_CHECK(!As.empty() && As.size() <= 4);
+  const long rows = B.size(0);
+  const int CA = As[0].size(1), N = B.size(1);
+  const int M = (int)As.size() * CA;
+  TORCH_CHECK(B.is_cuda() && B.is_contiguous() && B.dim() == 2);
+  TORCH_CHECK(M <= 256 && N <= 64 && CA % 8 == 0 && N % 8 == 0);
+  TORCH_CHECK(C.is_contiguous() && C.scalar_type() == at::kFloat &&
+              C.size(0) == M && C.size(1) == N);
+  const void* srcs[4];
+  for (size_t i = 0; i < As.size(); ++i) {
+    TORCH_CHECK(As[i].is_contiguous() && As[i].size(0) == rows &&
+                As[i].size(1) == CA &&
+                As[i].scalar_type() == B.scalar_type());
+    srcs[i] = As[i].data_ptr();
+  }
+  float* dbp = nullptr;
+  if (db.has_value()) {
+    TORCH_CHECK(db->is_contiguous() && db->scalar_type() == at::kFloat &&
+                db->numel() == N);
+    dbp = db->data_ptr<float>();
+  }
+  stmgcn_atb_wgrad_multi(stream(), dtype_code(B), srcs, (int)As.size(), CA,
+                         B.data_ptr(), C.data_ptr<float>(), dbp, rows, N);
 }
 
 at::Tensor mfma_probe(at::Tensor A, at::Tensor B) {
@@ -654,4 +703,6 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
         "out = alpha*(G @ xin) + beta*p1 on (B,N,C) (recurrence replay)");
   m.def("atb_wgrad_into", &atb_wgrad_into,
         "C += A^T B (+db += colsum B) into caller-provided fp32 buffers");
+  m.def("atb_wgrad_multi", &atb_wgrad_multi,
+        "All-support wgrad: C[k] += As[k]^T B in one launch, B read once");
 }

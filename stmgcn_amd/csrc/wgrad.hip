@@ -344,9 +344,18 @@ extern "C" void stmgcn_lstm_wgrad(void* stream_v, int dtype, const void* dA,
 // db[N] += colsum(B). M <= 256, N <= 64 (zero-padded to 16-multiples in LDS).
 // Serves the graph-conv dW = feat^T @ dZ and db = colsum(dZ) (SURVEY K2
 // backward) — measured 465 us each as hipBLASLt calls.
+//
+// Multi-source form: A is up to 4 separate (rows, CA) tensors laid out as
+// the column blocks of a virtual (rows, M = K*CA) matrix — the fused
+// ChebConv wgrad passes its recurrence states [x, p_1, .., p_{K_s-1}] so
+// ALL supports' dW_k land in one launch with dZ streamed ONCE (the
+// (B,N,K_s,C) concat stack never exists). CA must be a multiple of 8 so
+// every 8-col fragment stays within one source.
+struct AtbSrc { const void* a[4]; };
+
 template <typename T>
 __global__ void __launch_bounds__(256, 1)
-atb_wgrad_kernel(const T* __restrict__ A, const T* __restrict__ B,
+atb_wgrad_kernel(AtbSrc As, int CA, const T* __restrict__ B,
                  float* __restrict__ C, float* __restrict__ db,
                  long rows, int M, int N) {
   using frag = typename WFrag8<T>::type;
@@ -379,18 +388,24 @@ atb_wgrad_kernel(const T* __restrict__ A, const T* __restrict__ B,
   const int pr = threadIdx.x & 15;
   const int cba = threadIdx.x >> 4, cba2 = cba + 16;  // A col-blocks
   const int cbb = cba & 7;                            // B col-block (tid<128)
+  // resolve this thread's two col-blocks to (source tensor, column offset)
+  const T* Aa = nullptr;
+  const T* Ab = nullptr;
+  int oa = 0, ob = 0;
+  if (cba * 8 < M) { Aa = (const T*)As.a[(cba * 8) / CA]; oa = (cba * 8) % CA; }
+  if (cba2 * 8 < M) { Ab = (const T*)As.a[(cba2 * 8) / CA]; ob = (cba2 * 8) % CA; }
   const frag fz = {};
   frag va[2][2], vb[2];
   auto load_tiles = [&](long kt, frag a[2][2], frag b[2]) {
     const long ra = kt + pr * 2, rb = ra + 1;
     a[0][0] = fz; a[0][1] = fz; a[1][0] = fz; a[1][1] = fz;
-    if (cba * 8 < M) {
-      if (ra < r1) a[0][0] = *(const frag*)&A[ra * M + cba * 8];
-      if (rb < r1) a[0][1] = *(const frag*)&A[rb * M + cba * 8];
+    if (Aa != nullptr) {
+      if (ra < r1) a[0][0] = *(const frag*)&Aa[ra * CA + oa];
+      if (rb < r1) a[0][1] = *(const frag*)&Aa[rb * CA + oa];
     }
-    if (cba2 * 8 < M) {
-      if (ra < r1) a[1][0] = *(const frag*)&A[ra * M + cba2 * 8];
-      if (rb < r1) a[1][1] = *(const frag*)&A[rb * M + cba2 * 8];
+    if (Ab != nullptr) {
+      if (ra < r1) a[1][0] = *(const frag*)&Ab[ra * CA + ob];
+      if (rb < r1) a[1][1] = *(const frag*)&Ab[rb * CA + ob];
     }
     b[0] = fz; b[1] = fz;
     if (threadIdx.x < 128 && cbb * 8 < N) {
@@ -469,23 +484,33 @@ atb_wgrad_kernel(const T* __restrict__ A, const T* __restrict__ B,
   }
 }
 
-extern "C" void stmgcn_atb_wgrad(void* stream_v, int dtype, const void* A,
-                                 const void* B, float* C, float* db, long rows,
-                                 int M, int N) {
+extern "C" void stmgcn_atb_wgrad_multi(void* stream_v, int dtype,
+                                       const void** As, int nsrc, int CA,
+                                       const void* B, float* C, float* db,
+                                       long rows, int N) {
   // fill the 256-CU chip: ~128 rows per workgroup minimum (4 K-tiles)
   long nchunks = (rows + 127) / 128;
   if (nchunks > 512) nchunks = 512;
   if (nchunks < 1) nchunks = 1;
   const dim3 grid((unsigned)nchunks), blk(256);
   const size_t lds = 16384 + 4096;
+  const int M = nsrc * CA;
+  AtbSrc src{};
+  for (int i = 0; i < nsrc && i < 4; ++i) src.a[i] = As[i];
   hipStream_t stream = (hipStream_t)stream_v;
   if (dtype == STM_BF16)
     hipLaunchKernelGGL((atb_wgrad_kernel<__hip_bfloat16>), grid, blk, lds,
-                       stream, (const __hip_bfloat16*)A,
-                       (const __hip_bfloat16*)B, C, db, rows, M, N);
+                       stream, src, CA, (const __hip_bfloat16*)B, C, db, rows,
+                       M, N);
   else if (dtype == STM_F16)
     hipLaunchKernelGGL((atb_wgrad_kernel<__half>), grid, blk, lds, stream,
-                       (const __half*)A, (const __half*)B, C, db, rows, M, N);
+                       src, CA, (const __half*)B, C, db, rows, M, N);
   else
     printf("stmgcn_atb_wgrad: unsupported dtype %d (bf16/f16 only)\n", dtype);
+}
+
+extern "C" void stmgcn_atb_wgrad(void* stream_v, int dtype, const void* A,
+                                 const void* B, float* C, float* db, long rows,
+                                 int M, int N) {
+  stmgcn_atb_wgrad_multi(stream_v, dtype, &A, 1, M, B, C, db, rows, N);
 }

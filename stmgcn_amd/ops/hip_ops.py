@@ -80,20 +80,28 @@ class ChebGconvFn(torch.autograd.Function):
         Wd = W.to(x.dtype).contiguous()
         fused = (x.dtype in (torch.bfloat16, torch.float16)
                  and Cin <= 64 and Cout <= 64
-                 and Cin % 8 == 0 and Cout % 8 == 0)
+                 and Cin % 8 == 0 and Cout % 8 == 0
+                 and csr.K_supports <= 4)   # atb_wgrad_multi: <= 4 sources
         ctx.fused = fused
         ctx.csr = csr
         ctx.act = activation
         ctx.has_b = b is not None
         ctx.cin = Cin
         if fused:
-            y = C.cheb_gconv_fused_fwd(
+            training = torch.is_grad_enabled() and (
+                x.requires_grad or W.requires_grad
+                or (b is not None and b.requires_grad))
+            outs = C.cheb_gconv_fused_fwd(
                 x, csr.row_ptr, csr.col_idx, csr.vals, Wd,
                 b.to(x.dtype).contiguous() if b is not None else None,
                 csr.K_supports, csr.kind == "single",
-                1 if activation == "relu" else 0)
+                1 if activation == "relu" else 0, training)
+            y = outs[0]
+            # outs[1:] = recurrence states p_1..p_{K_s-1} (p_0 == x), kept
+            # for the wgrad — their HBM writes were mandatory regardless
             ctx.save_for_backward(x, Wd, W,
-                                  y if activation == "relu" else None)
+                                  y if activation == "relu" else None,
+                                  *outs[1:])
             return y
         S = C.cheb_apply(x, csr.row_ptr, csr.col_idx, csr.vals,
                          csr.K_supports, csr.kind == "single")
@@ -114,7 +122,8 @@ class ChebGconvFn(torch.autograd.Function):
         C = require_hip()
         csr: CSRSupport = ctx.csr
         if ctx.fused:
-            x, Wd, W, y = ctx.saved_tensors
+            x, Wd, W, y = ctx.saved_tensors[:4]
+            ps = ctx.saved_tensors[4:]       # p_1..p_{K_s-1} from forward
             if ctx.act == "relu":
                 dz = (dy * (y > 0).to(dy.dtype)).contiguous()
             else:
@@ -125,30 +134,16 @@ class ChebGconvFn(torch.autograd.Function):
             dX = C.cheb_gconv_fused_bwd_dx(dz, Wd, csr.row_ptr_t,
                                            csr.col_idx_t, csr.vals_t, K_s,
                                            single)
-            # ---- dW/db: replay the recurrence, one atb MFMA call per support
+            # ---- dW/db: ONE multi-source MFMA reduction over the saved
+            # recurrence states (dz streamed once, no concat stack)
             dWf = torch.zeros(K_s * Cin, Cout, dtype=torch.float32,
                               device=dz.device)
             db_f = (torch.zeros(Cout, dtype=torch.float32, device=dz.device)
                     if ctx.has_b else None)
             dz2 = dz.reshape(-1, Cout)
-            if single:
-                p = C.spmm_axpby(x, None, csr.row_ptr, csr.col_idx, csr.vals,
-                                 1.0, 0.0)
-                C.atb_wgrad_into(p.reshape(-1, Cin), dz2, dWf, db_f)
-            else:
-                C.atb_wgrad_into(x.reshape(-1, Cin), dz2, dWf[:Cin], db_f)
-                if K_s > 1:
-                    p1 = C.spmm_axpby(x, None, csr.row_ptr, csr.col_idx,
-                                      csr.vals, 1.0, 0.0)
-                    C.atb_wgrad_into(p1.reshape(-1, Cin), dz2,
-                                     dWf[Cin:2 * Cin], None)
-                    pm2, pm1 = x, p1
-                    for k in range(2, K_s):
-                        pk = C.spmm_axpby(pm1, pm2, csr.row_ptr, csr.col_idx,
-                                          csr.vals, 2.0, -1.0)
-                        C.atb_wgrad_into(pk.reshape(-1, Cin), dz2,
-                                         dWf[k * Cin:(k + 1) * Cin], None)
-                        pm2, pm1 = pm1, pk
+            srcs = [p.reshape(-1, Cin) for p in ps] if single \
+                else [x.reshape(-1, Cin)] + [p.reshape(-1, Cin) for p in ps]
+            C.atb_wgrad_multi(srcs, dz2, dWf, db_f)
             dW = dWf.to(W.dtype)
             db = db_f.to(W.dtype) if ctx.has_b else None
             return dX, dW, db, None, None

@@ -632,10 +632,16 @@ def test_trainer_hipgraph_capture_path():
             ls.append(tr._train_step(x[:8], y[:8], adjs))  # returns float
         losses[graph] = ls
         assert all(np.isfinite(v) for v in ls)
-    assert tr._graph is not None, "graph was never captured"
+    assert len(tr._graphs) == 1, "graph was never captured"
     # both runs converge; allow the extra capture-warmup steps' perturbation
     assert losses[True][-1] < 0.7 * losses[True][0]
     assert losses[False][-1] < 0.7 * losses[False][0]
+    # a ragged batch must NOT clobber the full-batch graph (per-shape cache)
+    full_key = next(iter(tr._graphs))
+    v = tr._train_step(x[:5], y[:5], adjs)   # eager warmup for the new shape
+    assert np.isfinite(v) and full_key in tr._graphs
+    v = tr._train_step(x[:8], y[:8], adjs)   # full shape still replays
+    assert np.isfinite(v)
 
 
 @pytest.mark.parametrize("dtype", [torch.bfloat16, torch.float16])
