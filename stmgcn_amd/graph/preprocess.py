@@ -46,12 +46,16 @@ def random_walk_normalize(adj: torch.Tensor) -> torch.Tensor:
     return adj * d_inv.unsqueeze(1)
 
 
-def power_iteration_lmax(L: torch.Tensor, iters: int = 100, tol: float = 1e-6) -> float:
-    """Largest eigenvalue of symmetric L by power iteration (the honest
-    replacement for the reference's dead torch.eig path)."""
+def power_iteration_lmax(L: torch.Tensor, iters: int = 500, tol: float = 1e-9) -> float:
+    """Largest eigenvalue of symmetric PSD L by power iteration (the honest
+    replacement for the reference's dead torch.eig path). Converges from
+    below; exit requires the Rayleigh quotient to stall for 3 consecutive
+    iterations (a single small delta is NOT convergence when the spectral
+    gap is tiny — the estimate can still be several percent short)."""
     v = torch.randn(L.shape[0], generator=torch.Generator().manual_seed(0), dtype=L.dtype)
     v = v / v.norm()
     lam = 0.0
+    stall = 0
     for _ in range(iters):
         w = L @ v
         nw = w.norm()
@@ -59,9 +63,10 @@ def power_iteration_lmax(L: torch.Tensor, iters: int = 100, tol: float = 1e-6) -
             return 0.0
         v_new = w / nw
         lam_new = float(v_new @ (L @ v_new))
-        if abs(lam_new - lam) < tol:
-            return lam_new
+        stall = stall + 1 if abs(lam_new - lam) < tol else 0
         lam, v = lam_new, v_new
+        if stall >= 3:
+            break
     return lam
 
 
@@ -171,7 +176,12 @@ class SupportGenerator:
             if self.lambda_max_mode == "fixed2":
                 lmax = 2.0  # the reference's only reachable path (quirk 1)
             elif self.lambda_max_mode == "power_iteration":
-                lmax = power_iteration_lmax(L)
+                # 1% spectral safety margin, clamped to the sym-normalized
+                # Laplacian's hard bound of 2: power iteration approaches
+                # lmax from BELOW, and an underestimate pushes the scaled
+                # spectrum outside [-1,1] (Chebyshev bound violated) while a
+                # slight overestimate merely compresses it.
+                lmax = min(power_iteration_lmax(L) * 1.01, 2.0)
             else:
                 raise ValueError(f"bad lambda_max_mode {self.lambda_max_mode!r}")
             return (2.0 / lmax) * L - torch.eye(L.shape[0])
