@@ -61,7 +61,13 @@ def gconv_mix(A, x: torch.Tensor, W: torch.Tensor, b: Optional[torch.Tensor],
     if isinstance(A, CSRSupport):
         if x.is_cuda and impl_mode() == "hip":
             from .hip_ops import ChebGconvFn
-            return ChebGconvFn.apply(x, W, b, A, activation)
+            # grad mode is disabled inside Function.forward, so the
+            # training decision (save recurrence states for wgrad?) must be
+            # made here
+            training = torch.is_grad_enabled() and (
+                x.requires_grad or W.requires_grad
+                or (b is not None and b.requires_grad))
+            return ChebGconvFn.apply(x, W, b, A, activation, training)
         return ref.gconv_mix_csr(A, x, W, b, activation)
     return ref.gconv_mix_dense(A, x, W, b, activation)
 

@@ -1,9 +1,10 @@
 """autograd.Function wrappers over the CDNA4 HIP kernels (stmgcn_amd._C).
 
 Kernel coverage (all landed; SURVEY §2.4 op numbers in parentheses):
-  - ChebGconvFn (K1/K2/K10): in-kernel CSR Chebyshev support recurrence +
-    hipBLASLt mix GEMM with bias epilogue; backward = Clenshaw over G^T and
-    the atb_wgrad reduction-GEMM kernel for dW/db.
+  - ChebGconvFn (K1/K2/K10): fully fused ChebConv — in-kernel CSR support
+    recurrence with the mix GEMM + bias + act as an MFMA epilogue of each
+    step; backward = Clenshaw over G^T with the U = dz W^T GEMM fused
+    in-kernel + one multi-source atb_wgrad launch for every dW_k.
   - FusedLSTMFn / FusedRNNFn (K5/K6/K10): persistent multi-layer LSTM/GRU
     (fused_rnn.hip) + one batched wgrad launch for ALL layer weight grads
     (wgrad.hip). GRU rides the 4-slot packing documented on FusedLSTMFn.
@@ -58,13 +59,14 @@ class ChebGconvFn(torch.autograd.Function):
     (reference GCN.py:34-42 materializes dense (K,N,N)
     stacks offline and a (B,N,K_s,C) feature concat per forward).
 
-    Fused path (bf16/f16, C/Cout <= 64 and mult-of-8 — every BASELINE
-    config): the K_s recurrence steps run in-kernel with the mix GEMM +
-    bias + ReLU folded into each step's MFMA epilogue; nothing but x and y
-    is saved. Backward: dX via Clenshaw over G^T with U_j = dz W_j^T fused
-    in-kernel; dW_k = (T_k x)^T dz by REPLAYING the recurrence (spmm_axpby)
-    and feeding each state to the atb_wgrad MFMA reduction kernel. Zero
-    library GEMMs, zero (B,N,K_s,C) stacks in either direction.
+    Fused path (bf16/f16, C/Cout <= 64 and mult-of-8, K_s <= 4 — every
+    BASELINE config): the K_s recurrence steps run in-kernel with the mix
+    GEMM + bias + ReLU folded into each step's MFMA epilogue. Training keeps
+    the recurrence states p_k (their HBM writes are mandatory anyway — the
+    step dependency crosses workgroups) and the backward computes dX via
+    Clenshaw over G^T with U_j = dz W_j^T fused in-kernel, plus ALL
+    supports' dW_k in one atb_wgrad_multi launch over [x, p_1, ..]. Zero
+    library GEMMs, zero (B,N,K_s,C) concat stacks in either direction.
 
     Stack path (fp32 parity / larger widths): cheb_apply support stack +
     rocBLAS mix, Clenshaw combine backward.
@@ -72,7 +74,8 @@ class ChebGconvFn(torch.autograd.Function):
 
     @staticmethod
     def forward(ctx, x: torch.Tensor, W: torch.Tensor, b: Optional[torch.Tensor],
-                csr: CSRSupport, activation: Optional[str]):
+                csr: CSRSupport, activation: Optional[str],
+                training: bool = True):
         C = require_hip()
         x = x.contiguous()
         Cin = x.shape[-1]
@@ -88,9 +91,6 @@ class ChebGconvFn(torch.autograd.Function):
         ctx.has_b = b is not None
         ctx.cin = Cin
         if fused:
-            training = torch.is_grad_enabled() and (
-                x.requires_grad or W.requires_grad
-                or (b is not None and b.requires_grad))
             outs = C.cheb_gconv_fused_fwd(
                 x, csr.row_ptr, csr.col_idx, csr.vals, Wd,
                 b.to(x.dtype).contiguous() if b is not None else None,
@@ -146,7 +146,7 @@ class ChebGconvFn(torch.autograd.Function):
             C.atb_wgrad_multi(srcs, dz2, dWf, db_f)
             dW = dWf.to(W.dtype)
             db = db_f.to(W.dtype) if ctx.has_b else None
-            return dX, dW, db, None, None
+            return dX, dW, db, None, None, None
 
         feat, W, y = ctx.saved_tensors
         if ctx.act == "relu":
@@ -172,7 +172,7 @@ class ChebGconvFn(torch.autograd.Function):
         U = (dz @ W.to(dz.dtype).T).view(B_, N, csr.K_supports, ctx.cin).contiguous()
         dX = C.cheb_combine(U, csr.row_ptr_t, csr.col_idx_t, csr.vals_t,
                             csr.kind == "single")
-        return dX, dW, db, None, None
+        return dX, dW, db, None, None, None
 
 
 def contextual_gate_hip(obs_seq, gconv_out, fc_weight, fc_bias):
