@@ -138,9 +138,10 @@ at::Tensor cheb_combine(at::Tensor U, at::Tensor rowptr, at::Tensor colidx,
 
 extern "C" void stmgcn_cheb_fused_fwd_step(
     void* stream, int dtype, const int* rowptr, const int* colidx,
-    const float* vals, const void* xin, const void* p1, const void* W,
-    const void* bias, void* pout, float* yacc, void* yout, int B, int N,
-    int C, int Cout, int kofs, float alpha, float beta, int first, int act);
+    const float* vals, const void* xin, const void* p1, const void* x0,
+    const void* W, const void* bias, void* pout, float* yacc, void* yout,
+    int B, int N, int C, int Cout, int kofs, float alpha, float beta,
+    int first, int act);
 extern "C" void stmgcn_cheb_fused_bwd_step(
     void* stream, int dtype, const int* rowptr, const int* colidx,
     const float* vals, const void* xin, const void* p1, const void* dz,
@@ -185,36 +186,40 @@ std::vector<at::Tensor> cheb_gconv_fused_fwd(
       pp = p.data_ptr();
     }
     stmgcn_cheb_fused_fwd_step(st, dt, rp, ci, v, x.data_ptr(), nullptr,
-                               W.data_ptr(), bp, pp, nullptr, y.data_ptr(),
-                               B, N, C, Cout, 0, 1.f, 0.f, 1, actc);
+                               nullptr, W.data_ptr(), bp, pp, nullptr,
+                               y.data_ptr(), B, N, C, Cout, 0, 1.f, 0.f, 1,
+                               actc);
     return training ? std::vector<at::Tensor>{y, p}
                     : std::vector<at::Tensor>{y};
   }
   if (K_s == 1) {  // T_0 only: y = act(x @ W + b)
     stmgcn_cheb_fused_fwd_step(st, dt, rp, ci, v, nullptr, x.data_ptr(),
-                               W.data_ptr(), bp, nullptr, nullptr,
+                               nullptr, W.data_ptr(), bp, nullptr, nullptr,
                                y.data_ptr(), B, N, C, Cout, 0, 0.f, 1.f, 1,
                                actc);
     return {y};
   }
-  auto yacc = at::empty({B, N, Cout}, x.options().dtype(at::kFloat));
-  float* ya = yacc.data_ptr<float>();
-  // k = 0: p_0 aliases x; mix only
-  stmgcn_cheb_fused_fwd_step(st, dt, rp, ci, v, nullptr, x.data_ptr(),
-                             W.data_ptr(), bp, nullptr, ya, nullptr, B, N, C,
-                             Cout, 0, 0.f, 1.f, 1, actc);
-  // k >= 1 recurrence states; training keeps every p_k (zero extra traffic —
-  // the writes are mandatory), eval ping/pongs two buffers.
+  // K_s - 1 launches: the T_0 mix (x @ W_0) rides the T_1 launch as its
+  // x0 operand, so K_s == 2 needs no fp32 accumulator at all.
+  at::Tensor yacc;
+  float* ya = nullptr;
+  if (K_s > 2) {
+    yacc = at::empty({B, N, Cout}, x.options().dtype(at::kFloat));
+    ya = yacc.data_ptr<float>();
+  }
+  // recurrence states; training keeps every p_k (zero extra traffic — the
+  // writes are mandatory), eval ping/pongs two buffers.
   std::vector<at::Tensor> ps;
   const int nbuf = training ? (int)K_s - 1 : std::min<int>(2, (int)K_s - 1);
   for (int i = 0; i < nbuf; ++i)
     ps.push_back(at::empty({B, N, C}, x.options()));
-  // k = 1: p_1 = G x
+  // k = 1: p_1 = G x; y (+)= x @ W_0 + p_1 @ W_1
   bool last = (K_s == 2);
   stmgcn_cheb_fused_fwd_step(st, dt, rp, ci, v, x.data_ptr(), nullptr,
-                             W.data_ptr(), bp, ps[0].data_ptr(), ya,
+                             x.data_ptr(), W.data_ptr(), bp,
+                             ps[0].data_ptr(), ya,
                              last ? y.data_ptr() : nullptr, B, N, C, Cout,
-                             C, 1.f, 0.f, 0, actc);
+                             C, 1.f, 0.f, 1, actc);
   // k >= 2: p_k = 2 G p_{k-1} - p_{k-2}; p1 == pout aliasing is element-safe
   const void* pm2 = x.data_ptr();
   void* pm1 = ps[0].data_ptr();
@@ -223,8 +228,9 @@ std::vector<at::Tensor> cheb_gconv_fused_fwd(
                          : (k == 2 ? ps[1].data_ptr()
                                    : const_cast<void*>(pm2));
     last = (k == K_s - 1);
-    stmgcn_cheb_fused_fwd_step(st, dt, rp, ci, v, pm1, pm2, W.data_ptr(), bp,
-                               dst, ya, last ? y.data_ptr() : nullptr, B, N,
+    stmgcn_cheb_fused_fwd_step(st, dt, rp, ci, v, pm1, pm2, nullptr,
+                               W.data_ptr(), bp, dst, ya,
+                               last ? y.data_ptr() : nullptr, B, N,
                                C, Cout, k * C, 2.f, -1.f, 0, actc);
     pm2 = pm1;
     pm1 = dst;
@@ -263,8 +269,9 @@ at::Tensor cheb_gconv_fused_bwd_dx(at::Tensor dz, at::Tensor W,
     auto U0 = at::empty({B, N, C}, dz.options());
     bwd(nullptr, nullptr, U0.data_ptr(), 0, 0.f, 0.f);
     stmgcn_cheb_fused_fwd_step(st, dt, rp, ci, v, U0.data_ptr(), nullptr,
-                               nullptr, nullptr, dX.data_ptr(), nullptr,
-                               nullptr, B, N, C, C, 0, 1.f, 0.f, 1, 0);
+                               nullptr, nullptr, nullptr, dX.data_ptr(),
+                               nullptr, nullptr, B, N, C, C, 0, 1.f, 0.f, 1,
+                               0);
     return dX;
   }
   const int K = (int)K_s - 1;
@@ -308,8 +315,8 @@ at::Tensor spmm_axpby(at::Tensor xin, c10::optional<at::Tensor> p1,
                              rowptr.data_ptr<int>(), colidx.data_ptr<int>(),
                              vals.data_ptr<float>(), xin.data_ptr(),
                              p1.has_value() ? p1->data_ptr() : nullptr,
-                             nullptr, nullptr, out.data_ptr(), nullptr,
-                             nullptr, B, N, C, C, 0, (float)alpha,
+                             nullptr, nullptr, nullptr, out.data_ptr(),
+                             nullptr, nullptr, B, N, C, C, 0, (float)alpha,
                              (float)beta, 1, 0);
   return out;
 }
@@ -531,6 +538,8 @@ void stmgcn_gate_bwd(void*, int, const void*, const void*, const void*,
 void stmgcn_head_fwd(void*, int, const void*, const void*, const void*,
                      const void*, const void*, void*, void*, int, long);
 void stmgcn_head_bwd(void*, int, const void*, const void*, void*, int, long);
+void stmgcn_head_wgrad(void*, int, const void*, const void*, float*, float*,
+                       int, long);
 void stmgcn_mse_fwd(void*, int, const void*, const void*, float*, void*, long);
 void stmgcn_mse_bwd(void*, int, const void*, const float*, void*, long);
 void stmgcn_adam(void*, int, float*, void*, const void*, float*, float*,
@@ -622,6 +631,20 @@ at::Tensor head_bwd(at::Tensor dy, at::Tensor w, int64_t G) {
   return dfeat;
 }
 
+// dw[g] = sum_r dy[r] fsum[r,g]; db = sum dy — fp32 accumulated reduction.
+std::vector<at::Tensor> head_wgrad(at::Tensor dy, at::Tensor fsum) {
+  dy = dy.contiguous();
+  fsum = fsum.contiguous();
+  const long BN = (long)fsum.size(0) * fsum.size(1);
+  const int G = fsum.size(2);
+  auto fopt = dy.options().dtype(at::kFloat);
+  auto dw = at::zeros({G}, fopt);
+  auto db = at::zeros({1}, fopt);
+  stmgcn_head_wgrad(stream(), dtype_code(dy), dy.data_ptr(), fsum.data_ptr(),
+                    dw.data_ptr<float>(), db.data_ptr<float>(), G, BN);
+  return {dw, db};
+}
+
 std::vector<at::Tensor> mse_fwd(at::Tensor pred, at::Tensor tgt) {
   pred = pred.contiguous();
   auto loss = at::zeros({}, pred.options().dtype(at::kFloat));
@@ -687,6 +710,7 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("gate_bwd", &gate_bwd, "K4: fused contextual gate backward");
   m.def("head_fwd", &head_fwd, "K7: branch-sum + FC head forward");
   m.def("head_bwd", &head_bwd, "K7: head backward (dfeat)");
+  m.def("head_wgrad", &head_wgrad, "K7: dw/db reduction (no library GEMM)");
   m.def("mse_fwd", &mse_fwd, "K8: fused MSE loss forward");
   m.def("mse_bwd", &mse_bwd, "K8: fused MSE grad");
   m.def("adam_step", &adam_step, "K9: multi-tensor Adam over flat arena");

@@ -145,6 +145,8 @@ namespace {
 //   tile = alpha * (G @ xin)[rows] + beta * p1[rows]     (either term optional)
 //   pout[rows] = tile                                    (optional)
 //   mix (W != null): yacc[rows] (+)= tile @ W[kofs:kofs+C, :]
+//                    (+ x0 != null: also += x0[rows] @ W[0:C, :] — the T_0
+//                     term folded into the T_1 launch, one launch fewer)
 //   final (yout != null): yout = act(yacc_total + bias)
 // p1 is read only at the thread's own (row, c) element, so p1 == pout
 // aliasing is safe (the gather operand xin must be a distinct buffer).
@@ -154,6 +156,7 @@ cheb_fused_fwd_kernel(const int* __restrict__ rowptr,
                       const int* __restrict__ colidx,
                       const float* __restrict__ vals,
                       const T* __restrict__ xin, const T* __restrict__ p1,
+                      const T* __restrict__ x0,
                       const T* __restrict__ W, const T* __restrict__ bias,
                       T* __restrict__ pout, float* __restrict__ yacc,
                       T* __restrict__ yout, int N, int C, int Cout, int kofs,
@@ -162,15 +165,23 @@ cheb_fused_fwd_kernel(const int* __restrict__ rowptr,
   const int n0 = blockIdx.x * CG_ST;
   extern __shared__ char lds[];
   float* fls = (float*)lds;
+  char* l0 = lds + (MF ? CG_ST * 128 : CG_ST * 64 * 4);  // x0 tile slot
+  float* fl0 = (float*)l0;
   const T* xb = xin ? xin + (long)b * N * C : nullptr;
   const T* pb = p1 ? p1 + (long)b * N * C : nullptr;
+  const T* x0b = x0 ? x0 + (long)b * N * C : nullptr;
 
   // ---- phase 1: recurrence (SpMM + axpby), tile staged in LDS ----
-  for (int i = threadIdx.x; i < CG_ST * 64; i += 256) {
-    const int c = i & 63, rl = i >> 6;
+  for (int i = threadIdx.x; i < CG_ST * C; i += 256) {
+    const int c = i % C, rl = i / C;
     const int row = n0 + rl;
     float acc = 0.f;
-    if (c < C && row < N) {
+    if (x0b != nullptr) {
+      const float v0 = (row < N) ? toF<T>(x0b[(long)row * C + c]) : 0.f;
+      if (MF) *(T*)&l0[lds_swz(rl, c * 2)] = fromF<T>(v0);
+      else fl0[rl * 64 + c] = v0;
+    }
+    if (row < N) {
       if (xb != nullptr && alpha != 0.f) {
         const int s = rowptr[row], e = rowptr[row + 1];
         int j = s;
@@ -221,6 +232,19 @@ cheb_fused_fwd_kernel(const int* __restrict__ rowptr,
         acc[0] = cg_mfma(a0, bfr, acc[0]);
         acc[1] = cg_mfma(a1, bfr, acc[1]);
       }
+      if (x0 != nullptr) {               // + x0 @ W_0 (weight rows [0, C))
+        for (int kk = 0; kk < kchunks; ++kk) {
+          frag bfr;
+          #pragma unroll
+          for (int j = 0; j < 8; ++j)
+            ((elem*)&bfr)[j] =
+                *(const elem*)&W[(kk * 32 + lgrp * 8 + j) * Cout + 16 * wv + l16];
+          frag a0 = *(const frag*)&l0[lds_swz(l16, (kk * 32 + lgrp * 8) * 2)];
+          frag a1 = *(const frag*)&l0[lds_swz(16 + l16, (kk * 32 + lgrp * 8) * 2)];
+          acc[0] = cg_mfma(a0, bfr, acc[0]);
+          acc[1] = cg_mfma(a1, bfr, acc[1]);
+        }
+      }
       #pragma unroll
       for (int m = 0; m < 2; ++m)
         #pragma unroll
@@ -248,6 +272,9 @@ cheb_fused_fwd_kernel(const int* __restrict__ rowptr,
       float v = 0.f;
       for (int c = 0; c < C; ++c)
         v = fmaf(fls[rl * 64 + c], toF<T>(W[(kofs + c) * Cout + co]), v);
+      if (x0 != nullptr)
+        for (int c = 0; c < C; ++c)
+          v = fmaf(fl0[rl * 64 + c], toF<T>(W[c * Cout + co]), v);
       const long idx = ((long)b * N + row) * Cout + co;
       if (!first) v += yacc[idx];
       if (yout != nullptr) {
@@ -283,10 +310,10 @@ cheb_fused_bwd_kernel(const int* __restrict__ rowptr,
 
   // ---- phase A: stage the dz tile ----
   const T* dzb = dz + (long)b * N * Cout;
-  for (int i = threadIdx.x; i < CG_ST * 64; i += 256) {
-    const int co = i & 63, rl = i >> 6;
+  for (int i = threadIdx.x; i < CG_ST * Cout; i += 256) {
+    const int co = i % Cout, rl = i / Cout;
     const int row = n0 + rl;
-    const float v = (co < Cout && row < N) ? toF<T>(dzb[(long)row * Cout + co]) : 0.f;
+    const float v = (row < N) ? toF<T>(dzb[(long)row * Cout + co]) : 0.f;
     if (MF) *(T*)&dzt[lds_swz(rl, co * 2)] = fromF<T>(v);
     else fdz[rl * 64 + co] = v;
   }
@@ -330,10 +357,10 @@ cheb_fused_bwd_kernel(const int* __restrict__ rowptr,
   // ---- phase C: out = alpha*(G^T @ xin) + beta*p1 + U ----
   const T* xb = xin ? xin + (long)b * N * C : nullptr;
   const T* pb = p1 ? p1 + (long)b * N * C : nullptr;
-  for (int i = threadIdx.x; i < CG_ST * 64; i += 256) {
-    const int c = i & 63, rl = i >> 6;
+  for (int i = threadIdx.x; i < CG_ST * C; i += 256) {
+    const int c = i % C, rl = i / C;
     const int row = n0 + rl;
-    if (c >= C || row >= N) continue;
+    if (row >= N) continue;
     float acc = ut[rl * 64 + c];
     if (xb != nullptr && alpha != 0.f) {
       float sp = 0.f;
@@ -361,24 +388,27 @@ cheb_fused_bwd_kernel(const int* __restrict__ rowptr,
 template <typename T>
 void launch_fused_fwd(hipStream_t st, const int* rp, const int* ci,
                       const float* v, const void* xin, const void* p1,
-                      const void* W, const void* bias, void* pout, float* yacc,
-                      void* yout, int B, int N, int C, int Cout, int kofs,
-                      float alpha, float beta, int first, int act) {
+                      const void* x0, const void* W, const void* bias,
+                      void* pout, float* yacc, void* yout, int B, int N,
+                      int C, int Cout, int kofs, float alpha, float beta,
+                      int first, int act) {
   dim3 grid((N + CG_ST - 1) / CG_ST, B);
+  const size_t nslot = x0 ? 2 : 1;
   if constexpr (!std::is_same<T, float>::value) {
     if (W != nullptr && (C % 32 == 0) && (Cout % 16 == 0)) {
       hipLaunchKernelGGL((cheb_fused_fwd_kernel<T, true>), grid, dim3(256),
-                         CG_ST * 128, st, rp, ci, v, (const T*)xin,
-                         (const T*)p1, (const T*)W, (const T*)bias, (T*)pout,
-                         yacc, (T*)yout, N, C, Cout, kofs, alpha, beta, first,
-                         act);
+                         nslot * CG_ST * 128, st, rp, ci, v, (const T*)xin,
+                         (const T*)p1, (const T*)x0, (const T*)W,
+                         (const T*)bias, (T*)pout, yacc, (T*)yout, N, C, Cout,
+                         kofs, alpha, beta, first, act);
       return;
     }
   }
   hipLaunchKernelGGL((cheb_fused_fwd_kernel<T, false>), grid, dim3(256),
-                     CG_ST * 64 * 4, st, rp, ci, v, (const T*)xin,
-                     (const T*)p1, (const T*)W, (const T*)bias, (T*)pout,
-                     yacc, (T*)yout, N, C, Cout, kofs, alpha, beta, first, act);
+                     nslot * CG_ST * 64 * 4, st, rp, ci, v, (const T*)xin,
+                     (const T*)p1, (const T*)x0, (const T*)W, (const T*)bias,
+                     (T*)pout, yacc, (T*)yout, N, C, Cout, kofs, alpha, beta,
+                     first, act);
 }
 
 template <typename T>
@@ -406,23 +436,24 @@ void launch_fused_bwd(hipStream_t st, const int* rp, const int* ci,
 
 extern "C" void stmgcn_cheb_fused_fwd_step(
     void* stream_v, int dtype, const int* rowptr, const int* colidx,
-    const float* vals, const void* xin, const void* p1, const void* W,
-    const void* bias, void* pout, float* yacc, void* yout, int B, int N,
-    int C, int Cout, int kofs, float alpha, float beta, int first, int act) {
+    const float* vals, const void* xin, const void* p1, const void* x0,
+    const void* W, const void* bias, void* pout, float* yacc, void* yout,
+    int B, int N, int C, int Cout, int kofs, float alpha, float beta,
+    int first, int act) {
   hipStream_t st = (hipStream_t)stream_v;
   switch (dtype) {
     case STM_F32:
-      launch_fused_fwd<float>(st, rowptr, colidx, vals, xin, p1, W, bias, pout,
-                              yacc, yout, B, N, C, Cout, kofs, alpha, beta,
-                              first, act);
+      launch_fused_fwd<float>(st, rowptr, colidx, vals, xin, p1, x0, W, bias,
+                              pout, yacc, yout, B, N, C, Cout, kofs, alpha,
+                              beta, first, act);
       break;
     case STM_BF16:
-      launch_fused_fwd<__hip_bfloat16>(st, rowptr, colidx, vals, xin, p1, W,
-                                       bias, pout, yacc, yout, B, N, C, Cout,
-                                       kofs, alpha, beta, first, act);
+      launch_fused_fwd<__hip_bfloat16>(st, rowptr, colidx, vals, xin, p1, x0,
+                                       W, bias, pout, yacc, yout, B, N, C,
+                                       Cout, kofs, alpha, beta, first, act);
       break;
     case STM_F16:
-      launch_fused_fwd<__half>(st, rowptr, colidx, vals, xin, p1, W, bias,
+      launch_fused_fwd<__half>(st, rowptr, colidx, vals, xin, p1, x0, W, bias,
                                pout, yacc, yout, B, N, C, Cout, kofs, alpha,
                                beta, first, act);
       break;

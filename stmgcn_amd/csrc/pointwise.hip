@@ -255,6 +255,39 @@ __global__ void head_bwd_kernel(const T* __restrict__ dy, const T* __restrict__ 
   dfeat[i] = fromF<T>(toF<T>(dy[i / G]) * toF<T>(w[i % G]));
 }
 
+// head wgrad: dw[g] = sum_r dy[r] * fsum[r,g]; db = sum_r dy[r] — a pure
+// bandwidth reduction (replaces the last library GEMM on the step: a
+// hipBLASLt MT64x16x512 tall-skinny at ~85 us for 1x64 output).
+template <typename T>
+__global__ void head_wgrad_kernel(const T* __restrict__ dy,
+                                  const T* __restrict__ fsum,
+                                  float* __restrict__ dw,   // (G,) zeroed
+                                  float* __restrict__ db,   // (1,) zeroed
+                                  int G, long BN) {
+  const int lane = threadIdx.x & 63;
+  const int wv = threadIdx.x >> 6;
+  const long chunk = (BN + gridDim.x - 1) / gridDim.x;
+  const long r0 = (long)blockIdx.x * chunk;
+  const long r1 = (r0 + chunk < BN) ? r0 + chunk : BN;
+  float acc = 0.f, accb = 0.f;
+  for (long r = r0 + wv; r < r1; r += 4) {
+    const float d = toF<T>(dy[r]);        // wave-uniform scalar load
+    if (lane < G) acc += d * toF<T>(fsum[r * G + lane]);
+    accb += d;
+  }
+  __shared__ float red[4][64];
+  __shared__ float redb[4];
+  red[wv][lane] = acc;
+  if (lane == 0) redb[wv] = accb;
+  __syncthreads();
+  if (wv == 0) {
+    const float v = red[0][lane] + red[1][lane] + red[2][lane] + red[3][lane];
+    if (lane < G) unsafeAtomicAdd(&dw[lane], v);
+    if (lane == 0)
+      unsafeAtomicAdd(db, redb[0] + redb[1] + redb[2] + redb[3]);
+  }
+}
+
 // ---- K8: fused MSE loss + grad -------------------------------------------
 template <typename T>
 __global__ void mse_fwd_kernel(const T* __restrict__ pred, const T* __restrict__ tgt,
@@ -454,6 +487,20 @@ void stmgcn_head_bwd(void* stream, int dtype, const void* dy, const void* w,
     case STM_F32: hipLaunchKernelGGL(head_bwd_kernel<float>, grid, dim3(256), 0, (hipStream_t)stream, (const float*)dy, (const float*)w, (float*)dfeat, G, total); break;
     case STM_BF16: hipLaunchKernelGGL(head_bwd_kernel<__hip_bfloat16>, grid, dim3(256), 0, (hipStream_t)stream, (const __hip_bfloat16*)dy, (const __hip_bfloat16*)w, (__hip_bfloat16*)dfeat, G, total); break;
     case STM_F16: hipLaunchKernelGGL(head_bwd_kernel<__half>, grid, dim3(256), 0, (hipStream_t)stream, (const __half*)dy, (const __half*)w, (__half*)dfeat, G, total); break;
+  }
+}
+
+void stmgcn_head_wgrad(void* stream, int dtype, const void* dy,
+                       const void* fsum, float* dw, float* db, int G,
+                       long BN) {
+  long nblk = (BN + 1023) / 1024;
+  if (nblk > 256) nblk = 256;
+  if (nblk < 1) nblk = 1;
+  dim3 grid((unsigned)nblk);
+  switch (dtype) {
+    case STM_F32: hipLaunchKernelGGL(head_wgrad_kernel<float>, grid, dim3(256), 0, (hipStream_t)stream, (const float*)dy, (const float*)fsum, dw, db, G, BN); break;
+    case STM_BF16: hipLaunchKernelGGL(head_wgrad_kernel<__hip_bfloat16>, grid, dim3(256), 0, (hipStream_t)stream, (const __hip_bfloat16*)dy, (const __hip_bfloat16*)fsum, dw, db, G, BN); break;
+    case STM_F16: hipLaunchKernelGGL(head_wgrad_kernel<__half>, grid, dim3(256), 0, (hipStream_t)stream, (const __half*)dy, (const __half*)fsum, dw, db, G, BN); break;
   }
 }
 

@@ -381,8 +381,11 @@ class HeadFn(torch.autograd.Function):
         fc_weight, fsum = ctx.saved_tensors
         G = fsum.shape[-1]
         dfeat = C.head_bwd(dy, fc_weight.view(-1).to(dy.dtype), G)
-        dw = (dy.reshape(1, -1) @ fsum.reshape(-1, G)).to(fc_weight.dtype)
-        db = dy.sum().reshape(1).to(fc_weight.dtype)
+        # dw = dy^T fsum, db = sum(dy) as one reduction kernel (the last
+        # library GEMM on the step was exactly this 1xG tall-skinny)
+        dwf, dbf = C.head_wgrad(dy, fsum)
+        dw = dwf.view(1, G).to(fc_weight.dtype)
+        db = dbf.to(fc_weight.dtype)
         return (dw, db) + (dfeat,) * ctx.M
 
 
