@@ -335,8 +335,8 @@ extern "C" void stmgcn_lstm_bwd(void* stream, int dtype, const void* dout,
                                 const void* x, const void* cseq_g,
                                 const void* gates_g, const void** w_ihT,
                                 const void** w_hhT, void* dx, void* dA_g,
-                                int S, int Tst, int L, int cin, int ret_seq,
-                                int gru);
+                                void* dh_g, int S, int Tst, int L, int cin,
+                                int ret_seq, int gru);
 extern "C" void stmgcn_mfma_probe(void* stream, const void* A, const void* B,
                                   void* D);
 extern "C" void stmgcn_lstm_wgrad(void* stream, int dtype, const void* dA,
@@ -383,12 +383,15 @@ std::vector<at::Tensor> lstm_fwd(at::Tensor x, std::vector<at::Tensor> w_ih,
     wi[l] = w_ih[l].data_ptr(); wh[l] = w_hh[l].data_ptr();
     bi[l] = b_ih[l].data_ptr(); bh[l] = b_hh[l].data_ptr();
   }
-  void *hp = nullptr, *cp = nullptr, *gp = nullptr;
+  // hseq is the cross-layer hand-off in GLOBAL memory (LDS holds only the
+  // live ping/pong slots) — allocated in eval too; cseq/gates only train.
+  hseq = at::empty({L, Tst, S_pad, kH}, x.options());
+  void* hp = hseq.data_ptr();
+  void *cp = nullptr, *gp = nullptr;
   if (training) {
-    hseq = at::empty({L, Tst, S_pad, kH}, x.options());
     cseq = at::empty({L, Tst, S_pad * kH}, x.options());
     gates = at::empty({L, Tst, S_pad * 4 * kH}, x.options());
-    hp = hseq.data_ptr(); cp = cseq.data_ptr(); gp = gates.data_ptr();
+    cp = cseq.data_ptr(); gp = gates.data_ptr();
   }
   stmgcn_lstm_fwd(stream(), dtype_code(x), x.data_ptr(), out.data_ptr(), hp,
                   cp, gp, wi, wh, bi, bh, S, Tst, L, cin, ret_seq ? 1 : 0,
@@ -410,6 +413,13 @@ std::vector<at::Tensor> lstm_bwd(at::Tensor dout, at::Tensor x,
   const long S_pad = nblk * kSeqTile;
   auto dx = at::empty_like(x);
   auto dA = at::empty({L, Tst, S_pad, 4 * kH}, x.options());
+  // cross-layer dh hand-off scratch (global; LDS only holds dA tiles)
+  at::Tensor dh;
+  void* dhp = nullptr;
+  if (L > 1) {
+    dh = at::empty({(long)Tst * S_pad * kH}, x.options());
+    dhp = dh.data_ptr();
+  }
   const void *wi[8], *wh[8];
   for (int l = 0; l < L; ++l) {
     TORCH_CHECK(w_ihT[l].is_contiguous() && w_hhT[l].is_contiguous());
@@ -417,7 +427,8 @@ std::vector<at::Tensor> lstm_bwd(at::Tensor dout, at::Tensor x,
   }
   stmgcn_lstm_bwd(stream(), dtype_code(x), dout.data_ptr(), x.data_ptr(),
                   cseq.data_ptr(), gates.data_ptr(), wi, wh, dx.data_ptr(),
-                  dA.data_ptr(), S, Tst, L, cin, ret_seq ? 1 : 0, gru ? 1 : 0);
+                  dA.data_ptr(), dhp, S, Tst, L, cin, ret_seq ? 1 : 0,
+                  gru ? 1 : 0);
   return {dx, dA};
 }
 

@@ -99,13 +99,19 @@ extern "C" __global__ void mfma_probe_kernel(const __hip_bfloat16* A,
 // save holds h_{t-1} (GRU) instead of c_t (LSTM). Everything else — LDS
 // staging, MFMA tiling, dA streaming, the batched wgrad kernel — is shared.
 //
-// LDS map (dynamic): hseq slots [T_steps][64][64] T-typed, swizzled;
-// then xbuf: CIN1 ? [T_steps][64] : [T_steps][64][64] (swizzled).
+// LDS holds only the LIVE state: two ping/pong h slots (t-1 and t of the
+// CURRENT layer, parity-indexed). The cross-layer sequence hand-off goes
+// through hseq_g in GLOBAL memory — in training those writes were already
+// mandatory (the wgrad kernel consumes hseq_g), so the hand-off is free and
+// LDS drops from (2*Tst+eps) slots (64 KB at T=8 -> 2 workgroups/CU) to
+// 2 slots (~8.5 KB -> occupancy is register-bound at 4 waves/SIMD instead).
+// CIN1 additionally stages the scalar input sequence ([Tst][ST] T).
 template <typename T, bool CIN1, bool GRU, int ST>
-__global__ void __launch_bounds__(256, 1)
+__global__ void __launch_bounds__(256, 3)
 lstm_fwd_kernel(const T* __restrict__ x,    // (S, Tst, C_in)
                 T* __restrict__ out,        // (S, H) or (S, Tst, H)
-                T* __restrict__ hseq_g,     // (L, Tst, S_pad, H) or null
+                T* __restrict__ hseq_g,     // (L, Tst, S_pad, H) — ALWAYS
+                                            // (layer hand-off + wgrad save)
                 T* __restrict__ cseq_g,     // (L, Tst, S_pad*H) frag-native
                                             // (LSTM c_t / GRU h_{t-1})
                 T* __restrict__ gates_g,    // (L, Tst, S_pad*4H) frag-native
@@ -122,45 +128,25 @@ lstm_fwd_kernel(const T* __restrict__ x,    // (S, Tst, C_in)
   const int lgrp = lane >> 4;
   const long S_pad = (long)gridDim.x * ST;
 
-  // CIN1: ping/pong h buffers by LAYER parity — the writer never aliases
-  // the prev-layer slots being read, so the pre-write WAR barrier drops
-  // (one barrier per (layer,t) stage instead of two). Dense-input variants
-  // keep the single in-place buffer (a third T*SLOT buffer would push LDS
-  // past 2-WG occupancy).
-  char* hseq = lds;                                   // Tst * SLOT bytes
-  char* hpong = lds + (CIN1 ? Tst * SLOT : 0);        // CIN1 only
-  char* xbuf = lds + (CIN1 ? 2 : 1) * (long)Tst * SLOT;
+  // ping/pong h slots by t parity (explicit ternary: an LDS-pointer array
+  // initializer is rejected by the AMDGPU backend)
+  auto hslot = [&](int par) -> char* { return lds + (par & 1) * SLOT; };
+  char* xbuf = lds + 2 * SLOT;                 // CIN1 scalar input stage
 
-  // ---- stage input x into LDS --------------------------------------------
+  // ---- stage scalar input x into LDS (CIN1 only) -------------------------
   if (CIN1) {
-    // x (S, Tst, 1) -> xbuf[t*ST + s] (T-typed)
     for (int i = threadIdx.x; i < ST * Tst; i += 256) {
-      const int s = i & (ST - 1), t = i / ST;
+      const int s = i % ST, t = i / ST;
       T v = fromF<T>(0.f);
       if (s0 + s < S) v = x[(long)(s0 + s) * Tst + t];
       ((T*)xbuf)[t * ST + s] = v;
     }
-  } else {
-    // x (S, Tst, 64) -> xbuf slot t swizzled, 16B pieces
-    for (int i = threadIdx.x; i < ST * Tst * 8; i += 256) {
-      const int c8 = i & 7, s = (i >> 3) & (ST - 1), t = (i >> 3) / ST;
-      frag v = {};
-      if (s0 + s < S)
-        v = *(const frag*)&x[((long)(s0 + s) * Tst + t) * RNN_H + c8 * 8];
-      *(frag*)&xbuf[t * SLOT + lds_swz(s, c8 * 16)] = v;
-    }
+    __syncthreads();
   }
-  __syncthreads();
 
-  // per-lane fragment row/col constants
-  //   acc row r(m, reg) = 16m + 4*lgrp + reg ; acc col = l16 (+16w for h ch)
   const int hch = 16 * wv + l16;                      // this lane's h channel
 
   for (int layer = 0; layer < L; ++layer) {
-    const int cin = (layer == 0) ? (CIN1 ? 1 : RNN_H) : RNN_H;
-    // this layer's write buffer / previous layer's read buffer
-    char* wrbuf = CIN1 ? ((layer & 1) ? hseq : hpong) : hseq;
-    char* rdbuf = CIN1 ? ((layer & 1) ? hpong : hseq) : hseq;
     const T* Whh = (const T*)ptrs.w_hh[layer];
     const T* Wih = (const T*)ptrs.w_ih[layer];
     const T* bih = (const T*)ptrs.b_ih[layer];
@@ -170,7 +156,7 @@ lstm_fwd_kernel(const T* __restrict__ x,    // (S, Tst, C_in)
     // cols kk*32 + lgrp*8 .. +8 -> contiguous 16B in the (4H, H) weight.
     frag bhfrag[4][2];
     #pragma unroll
-      for (int q = 0; q < 4; ++q)
+    for (int q = 0; q < 4; ++q)
       #pragma unroll
       for (int kk = 0; kk < 2; ++kk)
         bhfrag[q][kk] = *(const frag*)&Whh[(q * 64 + hch) * RNN_H + kk * 32 + lgrp * 8];
@@ -180,21 +166,26 @@ lstm_fwd_kernel(const T* __restrict__ x,    // (S, Tst, C_in)
       #pragma unroll
       for (int q = 0; q < 4; ++q)
         #pragma unroll
-      for (int kk = 0; kk < 2; ++kk)
+        for (int kk = 0; kk < 2; ++kk)
           bxfrag[q][kk] = *(const frag*)&Wih[(q * 64 + hch) * RNN_H + kk * 32 + lgrp * 8];
     // layer-0 scalar input weights (CIN1): W_ih[g][0]
     float wih0[4];
     float bias[4];
     #pragma unroll
-      for (int q = 0; q < 4; ++q) {
+    for (int q = 0; q < 4; ++q) {
       const int g = q * 64 + hch;
       bias[q] = toF<T>(bih[g]) + toF<T>(bhh[g]);
       if (CIN1 && layer == 0) wih0[q] = toF<T>(Wih[g]);
     }
 
+    // previous layer's output sequence (global hand-off)
+    const T* hin = (layer > 0)
+        ? hseq_g + ((long)(layer - 1) * Tst) * (S_pad * RNN_H)
+        : nullptr;
+
     float c_state[MT][4];  // [m][reg]
     #pragma unroll
-      for (int m = 0; m < MT; ++m)
+    for (int m = 0; m < MT; ++m)
       #pragma unroll
       for (int r = 0; r < 4; ++r) c_state[m][r] = 0.f;
 
@@ -203,35 +194,42 @@ lstm_fwd_kernel(const T* __restrict__ x,    // (S, Tst, C_in)
       #pragma unroll
       for (int m = 0; m < MT; ++m)
         #pragma unroll
-      for (int q = 0; q < 4; ++q) acc[m][q] = f32x4{0.f, 0.f, 0.f, 0.f};
+        for (int q = 0; q < 4; ++q) acc[m][q] = f32x4{0.f, 0.f, 0.f, 0.f};
 
-      // recurrent term: h_{t-1} from this layer's write buffer (zero at t==0)
+      // recurrent term: h_{t-1} from the previous parity slot (zero at t==0)
       if (t > 0) {
-        char* slot = wrbuf + (t - 1) * SLOT;
+        char* slot = hslot(t - 1);
         #pragma unroll
-      for (int m = 0; m < MT; ++m) {
+        for (int m = 0; m < MT; ++m) {
           const int row = 16 * m + l16;
           #pragma unroll
-      for (int kk = 0; kk < 2; ++kk) {
+          for (int kk = 0; kk < 2; ++kk) {
             frag a = *(const frag*)&slot[lds_swz(row, (kk * 32 + lgrp * 8) * 2)];
             #pragma unroll
-      for (int q = 0; q < 4; ++q)
+            for (int q = 0; q < 4; ++q)
               acc[m][q] = mfma16x16x32(a, bhfrag[q][kk], acc[m][q]);
           }
         }
       }
-      // input term: previous layer's h (dense layer 0: the staged x)
+      // input term: previous layer's h from GLOBAL (L2-hot: this block wrote
+      // the same tile last layer); dense layer 0 reads x directly
       if (!CIN1 || layer > 0) {
-        char* src = (!CIN1 && layer == 0) ? (xbuf + t * SLOT)
-                                          : (rdbuf + t * SLOT);
         #pragma unroll
-      for (int m = 0; m < MT; ++m) {
+        for (int m = 0; m < MT; ++m) {
           const int row = 16 * m + l16;
           #pragma unroll
-      for (int kk = 0; kk < 2; ++kk) {
-            frag a = *(const frag*)&src[lds_swz(row, (kk * 32 + lgrp * 8) * 2)];
+          for (int kk = 0; kk < 2; ++kk) {
+            frag a;
+            if (layer == 0) {          // dense x (S, Tst, 64): guard the tail
+              if (s0 + row < S)
+                a = *(const frag*)&x[((long)(s0 + row) * Tst + t) * RNN_H + kk * 32 + lgrp * 8];
+              else
+                a = frag{};
+            } else {
+              a = *(const frag*)&hin[((long)t * S_pad + s0 + row) * RNN_H + kk * 32 + lgrp * 8];
+            }
             #pragma unroll
-      for (int q = 0; q < 4; ++q)
+            for (int q = 0; q < 4; ++q)
               acc[m][q] = mfma16x16x32(a, bxfrag[q][kk], acc[m][q]);
           }
         }
@@ -246,10 +244,10 @@ lstm_fwd_kernel(const T* __restrict__ x,    // (S, Tst, C_in)
         float xv[4];
         if (CIN1 && layer == 0)
           #pragma unroll
-      for (int r = 0; r < 4; ++r)
+          for (int r = 0; r < 4; ++r)
             xv[r] = toF<T>(((const T*)xbuf)[t * ST + 16 * m + 4 * lgrp + r]);
         #pragma unroll
-      for (int r = 0; r < 4; ++r) {
+        for (int r = 0; r < 4; ++r) {
           float gi = acc[m][0][r] + bias[0];
           float gf = acc[m][1][r] + bias[1];
           float gg = acc[m][2][r] + bias[2];
@@ -287,20 +285,37 @@ lstm_fwd_kernel(const T* __restrict__ x,    // (S, Tst, C_in)
         }
       }
 
-      // dense in-place variant: all waves must finish reading slot t
-      // (input term) before it is overwritten; ping/pong needs no WAR wait
-      if (!CIN1) __syncthreads();
+      // write h_t into the parity slot: stage t-1's readers of this slot
+      // (= slot (t-2)&1 == t&1) finished before the previous stage barrier
       {
-        char* slot = wrbuf + t * SLOT;
+        char* slot = hslot(t);
         #pragma unroll
-      for (int m = 0; m < MT; ++m)
+        for (int m = 0; m < MT; ++m)
           #pragma unroll
-      for (int r = 0; r < 4; ++r) {
+          for (int r = 0; r < 4; ++r) {
             const int row = 16 * m + 4 * lgrp + r;
             *(T*)&slot[lds_swz(row, hch * 2)] = hval[m][r];
           }
       }
       __syncthreads();
+
+      // hand the slot off to global (next layer's input + wgrad/out source)
+      {
+        T* hp = hseq_g + ((long)layer * Tst + t) * (S_pad * RNN_H)
+                + (long)s0 * RNN_H;
+        char* slot = hslot(t);
+        for (int i = threadIdx.x; i < ST * 8; i += 256) {
+          const int c8 = i & 7, sr = i >> 3;
+          frag v = *(frag*)&slot[lds_swz(sr, c8 * 16)];
+          *(frag*)&hp[sr * RNN_H + c8 * 8] = v;
+          if (layer == L - 1 && s0 + sr < S) {
+            if (ret_seq)
+              *(frag*)&out[((long)(s0 + sr) * Tst + t) * RNN_H + c8 * 8] = v;
+            else if (t == Tst - 1)
+              *(frag*)&out[(long)(s0 + sr) * RNN_H + c8 * 8] = v;
+          }
+        }
+      }
 
       // training saves: gates + cell, fragment-native (16B contiguous/lane)
       if (gates_g) {
@@ -310,7 +325,7 @@ lstm_fwd_kernel(const T* __restrict__ x,    // (S, Tst, C_in)
                 + (long)blockIdx.x * (ST * 4 * RNN_H)
                 + ((wv * MT) * 64) * 16;
         #pragma unroll
-      for (int m = 0; m < MT; ++m)
+        for (int m = 0; m < MT; ++m)
           *(((frag*)(gp + (m * 64 + lane) * 16)) + 0) = *(frag*)&gsave[m][0],
           *(((frag*)(gp + (m * 64 + lane) * 16)) + 1) = *(frag*)&gsave[m][8];
         // cell: (L,Tst, S_pad*H) model-dtype as [wave][m][lane][4]
@@ -319,42 +334,18 @@ lstm_fwd_kernel(const T* __restrict__ x,    // (S, Tst, C_in)
         T* cp = cseq_g + base * (S_pad * RNN_H)
                 + (long)blockIdx.x * (ST * RNN_H) + (wv * MT) * 64 * 4;
         #pragma unroll
-      for (int m = 0; m < MT; ++m) {
+        for (int m = 0; m < MT; ++m) {
           T c4[4];
           #pragma unroll
           for (int r = 0; r < 4; ++r) c4[r] = fromF<T>(csave[m][r]);
           *(ulong1*)(cp + (m * 64 + lane) * 4) = *(ulong1*)c4;
         }
-        // hseq natural layout copy of slot t (also next layer's input source)
-        // (no trailing barrier: the next write to this address is at least
-        // one barrier away in both variants)
-        T* hp = hseq_g + base * (S_pad * RNN_H) + (long)(s0)*RNN_H;
-        char* slot = wrbuf + t * SLOT;
-        for (int i = threadIdx.x; i < ST * 8; i += 256) {
-          const int c8 = i & 7, s = i >> 3;
-          *(frag*)&hp[s * RNN_H + c8 * 8] = *(frag*)&slot[lds_swz(s, c8 * 16)];
-        }
       }
     }
-  }
-
-  // ---- output (from the LAST layer's write buffer) -----------------------
-  char* lastbuf = CIN1 ? (((L - 1) & 1) ? hseq : hpong) : hseq;
-  if (ret_seq) {
-    for (int i = threadIdx.x; i < ST * Tst * 8; i += 256) {
-      const int c8 = i & 7, s = (i >> 3) & (ST - 1), t = (i >> 3) / ST;
-      if (s0 + s < S)
-        *(frag*)&out[((long)(s0 + s) * Tst + t) * RNN_H + c8 * 8] =
-            *(frag*)&lastbuf[t * SLOT + lds_swz(s, c8 * 16)];
-    }
-  } else {
-    char* slot = lastbuf + (Tst - 1) * SLOT;
-    for (int i = threadIdx.x; i < ST * 8; i += 256) {
-      const int c8 = i & 7, s = i >> 3;
-      if (s0 + s < S)
-        *(frag*)&out[(long)(s0 + s) * RNN_H + c8 * 8] =
-            *(frag*)&slot[lds_swz(s, c8 * 16)];
-    }
+    // layer boundary: the hseq_g stores must be visible to this block's
+    // next-layer input loads (workgroup-scope fence + barrier)
+    __threadfence_block();
+    __syncthreads();
   }
 }
 
@@ -366,9 +357,7 @@ void launch_fwd(hipStream_t stream, const void* x, void* out, void* hseq_g,
   const int nblk = (S + ST - 1) / ST;
   const bool cin1 = (cin == 1);
   const size_t slot = (size_t)ST * 128;
-  const size_t lds_bytes = cin1
-      ? 2 * (size_t)Tst * slot + Tst * ST * sizeof(T)   // ping/pong + x scalars
-      : 2 * (size_t)Tst * slot;                         // in-place h + dense x
+  const size_t lds_bytes = 2 * slot + (cin1 ? Tst * ST * sizeof(T) : 0);
   auto go = [&](auto kern) {
     hipLaunchKernelGGL(kern, dim3(nblk), dim3(256), lds_bytes, stream,
                        (const T*)x, (T*)out, (T*)hseq_g, (T*)cseq_g,
@@ -430,7 +419,7 @@ __device__ __forceinline__ int swzA(int s, int cbyte) {      // dA rows: 512 B
 // transposed weights and the batched wgrad kernel run unchanged. The direct
 // dh_{t-1} += dh_t * z term rides the dc[][] register carry.
 template <typename T, bool CIN1, bool GRU, int ST>
-__global__ void __launch_bounds__(256, 1)
+__global__ void __launch_bounds__(256, 3)
 lstm_bwd_kernel(const T* __restrict__ dout,     // (S,H) or (S,Tst,H)
                 const T* __restrict__ x,        // (S,Tst,Cin)
                 const T* __restrict__ cseq_g,   // model dtype (see fwd)
@@ -438,11 +427,12 @@ lstm_bwd_kernel(const T* __restrict__ dout,     // (S,H) or (S,Tst,H)
                 RnnPtrs w,                      // w_ih/w_hh = TRANSPOSED (C|H, 4H)
                 T* __restrict__ dx,             // (S,Tst,Cin)
                 T* __restrict__ dA_g,           // (L,Tst,S_pad,4H)
+                T* __restrict__ dh_g,           // (Tst,S_pad,H) layer hand-off
+                                                // scratch (null when L == 1)
                 int S, int Tst, int L, int ret_seq) {
   using frag = typename Frag8<T>::type;
   using elem = typename Frag8<T>::elem;
   constexpr int MT = ST / 16;
-  constexpr int SLOT = ST * 128;
   extern __shared__ char lds[];
   const int s0 = blockIdx.x * ST;
   const int wv = threadIdx.x >> 6;
@@ -452,11 +442,13 @@ lstm_bwd_kernel(const T* __restrict__ dout,     // (S,H) or (S,Tst,H)
   const long S_pad = (long)gridDim.x * ST;
   const int hch = 16 * wv + l16;
 
-  char* dh_lds = lds;                         // [Tst][ST][64] fwd-slot layout
-  // dA tiles ping/pong by timestep parity: the pointwise writes of step t
-  // never alias the buffer step t+1's GEMMs/stream read, so the loop-top
-  // WAR barrier drops (one barrier per stage; bwd was 70% wave-parked)
-  char* dA_buf0 = lds + Tst * SLOT;           // [ST][512B] swizzled
+  // LDS holds only the dA tiles (ping/pong by timestep parity: the pointwise
+  // writes of step t never alias the buffer step t+1's GEMMs/stream read, so
+  // the loop-top WAR barrier drops) + the CIN1 dx reduction scratch. The
+  // cross-layer dh hand-off lives in GLOBAL scratch dh_g — dropping the
+  // [Tst] dh slots cuts LDS from ~64.5 KB (2 WGs/CU) to ~32.5 KB (4 WGs/CU),
+  // doubling the occupancy that hides the serial BPTT latency.
+  char* dA_buf0 = lds;                        // [ST][512B] swizzled
   char* dA_buf1 = dA_buf0 + ST * 512;
   float* red = (float*)(dA_buf1 + ST * 512);  // [4][ST] cross-wave scratch
 
@@ -488,8 +480,9 @@ lstm_bwd_kernel(const T* __restrict__ dout,     // (S,H) or (S,Tst,H)
       gcur[m][0] = *(((const frag*)(g_at(Tst - 1) + (m * 64 + lane) * 16)) + 0);
       gcur[m][1] = *(((const frag*)(g_at(Tst - 1) + (m * 64 + lane) * 16)) + 1);
     }
-    // layer boundary: for odd Tst the first stage reuses the buffer parity
-    // of the previous layer's last stage — its readers must be done
+    // layer boundary: dA-buffer parity reuse needs the readers done, and
+    // this block's dh_g stores must be visible to its next-layer loads
+    __threadfence_block();
     __syncthreads();
 
     for (int t = Tst - 1; t >= 0; --t) {
@@ -524,7 +517,7 @@ lstm_bwd_kernel(const T* __restrict__ dout,     // (S,H) or (S,Tst,H)
           const int row = 16 * m + 4 * lgrp + r;
           float dh = dh_rec[m][r];
           if (layer < L - 1) {
-            dh += toF<T>(*(const T*)&dh_lds[t * SLOT + lds_swz(row, hch * 2)]);
+            dh += toF<T>(dh_g[((long)t * S_pad + s0 + row) * RNN_H + hch]);
           } else if (ret_seq) {
             if (s0 + row < S) dh += toF<T>(dout[((long)(s0 + row) * Tst + t) * RNN_H + hch]);
           } else if (t == Tst - 1) {
@@ -620,7 +613,7 @@ lstm_bwd_kernel(const T* __restrict__ dout,     // (S,H) or (S,Tst,H)
             #pragma unroll
             for (int r = 0; r < 4; ++r) {
               const int row = 16 * m + 4 * lgrp + r;
-              *(T*)&dh_lds[t * SLOT + lds_swz(row, hch * 2)] = fromF<T>(acc[m][r]);
+              dh_g[((long)t * S_pad + s0 + row) * RNN_H + hch] = fromF<T>(acc[m][r]);
             }
         } else {
           #pragma unroll
@@ -676,16 +669,16 @@ lstm_bwd_kernel(const T* __restrict__ dout,     // (S,H) or (S,Tst,H)
 template <typename T>
 void launch_bwd(hipStream_t stream, const void* dout, const void* x,
                 const void* cseq_g, const void* gates_g, const RnnPtrs& w,
-                void* dx, void* dA_g, int S, int Tst, int L, int cin,
-                int ret_seq, int gru) {
+                void* dx, void* dA_g, void* dh_g, int S, int Tst, int L,
+                int cin, int ret_seq, int gru) {
   constexpr int ST = SEQ_TILE;
   const int nblk = (S + ST - 1) / ST;
-  const size_t lds_bytes = (size_t)Tst * ST * 128 + 2 * ST * 512 + 4 * ST * sizeof(float);
+  const size_t lds_bytes = 2 * ST * 512 + 4 * ST * sizeof(float);
   auto go = [&](auto kern) {
     hipLaunchKernelGGL(kern, dim3(nblk), dim3(256), lds_bytes, stream,
                        (const T*)dout, (const T*)x, (const T*)cseq_g,
-                       (const T*)gates_g, w, (T*)dx, (T*)dA_g, S, Tst, L,
-                       ret_seq);
+                       (const T*)gates_g, w, (T*)dx, (T*)dA_g, (T*)dh_g,
+                       S, Tst, L, ret_seq);
   };
   if (cin == 1 && !gru) go(lstm_bwd_kernel<T, true, false, ST>);
   else if (cin == 1 && gru) go(lstm_bwd_kernel<T, true, true, ST>);
@@ -697,8 +690,8 @@ extern "C" void stmgcn_lstm_bwd(void* stream_v, int dtype, const void* dout,
                                 const void* x, const void* cseq_g,
                                 const void* gates_g, const void** w_ihT,
                                 const void** w_hhT, void* dx, void* dA_g,
-                                int S, int Tst, int L, int cin, int ret_seq,
-                                int gru) {
+                                void* dh_g, int S, int Tst, int L, int cin,
+                                int ret_seq, int gru) {
   RnnPtrs p;
   for (int l = 0; l < L && l < MAX_LAYERS; ++l) {
     p.w_ih[l] = w_ihT[l]; p.w_hh[l] = w_hhT[l];
@@ -707,8 +700,8 @@ extern "C" void stmgcn_lstm_bwd(void* stream_v, int dtype, const void* dout,
   hipStream_t stream = (hipStream_t)stream_v;
   if (dtype == STM_BF16)
     launch_bwd<__hip_bfloat16>(stream, dout, x, cseq_g, gates_g, p, dx, dA_g,
-                               S, Tst, L, cin, ret_seq, gru);
+                               dh_g, S, Tst, L, cin, ret_seq, gru);
   else if (dtype == STM_F16)
-    launch_bwd<__half>(stream, dout, x, cseq_g, gates_g, p, dx, dA_g, S, Tst,
-                       L, cin, ret_seq, gru);
+    launch_bwd<__half>(stream, dout, x, cseq_g, gates_g, p, dx, dA_g, dh_g,
+                       S, Tst, L, cin, ret_seq, gru);
 }
