@@ -107,7 +107,7 @@ extern "C" __global__ void mfma_probe_kernel(const __hip_bfloat16* A,
 // 2 slots (~8.5 KB -> occupancy is register-bound at 4 waves/SIMD instead).
 // CIN1 additionally stages the scalar input sequence ([Tst][ST] T).
 template <typename T, bool CIN1, bool GRU, int ST>
-__global__ void __launch_bounds__(256, 3)
+__global__ void __launch_bounds__(256, 2)
 lstm_fwd_kernel(const T* __restrict__ x,    // (S, Tst, C_in)
                 T* __restrict__ out,        // (S, H) or (S, Tst, H)
                 T* __restrict__ hseq_g,     // (L, Tst, S_pad, H) — ALWAYS
@@ -299,15 +299,17 @@ lstm_fwd_kernel(const T* __restrict__ x,    // (S, Tst, C_in)
       }
       __syncthreads();
 
-      // hand the slot off to global (next layer's input + wgrad/out source)
+      // hand the slot off to global (next layer's input + wgrad/out source;
+      // the last layer's sequence is dead in eval — only `out` is written)
       {
         T* hp = hseq_g + ((long)layer * Tst + t) * (S_pad * RNN_H)
                 + (long)s0 * RNN_H;
+        const bool keep_h = (layer < L - 1) || (gates_g != nullptr);
         char* slot = hslot(t);
         for (int i = threadIdx.x; i < ST * 8; i += 256) {
           const int c8 = i & 7, sr = i >> 3;
           frag v = *(frag*)&slot[lds_swz(sr, c8 * 16)];
-          *(frag*)&hp[sr * RNN_H + c8 * 8] = v;
+          if (keep_h) *(frag*)&hp[sr * RNN_H + c8 * 8] = v;
           if (layer == L - 1 && s0 + sr < S) {
             if (ret_seq)
               *(frag*)&out[((long)(s0 + sr) * Tst + t) * RNN_H + c8 * 8] = v;
@@ -419,7 +421,7 @@ __device__ __forceinline__ int swzA(int s, int cbyte) {      // dA rows: 512 B
 // transposed weights and the batched wgrad kernel run unchanged. The direct
 // dh_{t-1} += dh_t * z term rides the dc[][] register carry.
 template <typename T, bool CIN1, bool GRU, int ST>
-__global__ void __launch_bounds__(256, 3)
+__global__ void __launch_bounds__(256, 2)
 lstm_bwd_kernel(const T* __restrict__ dout,     // (S,H) or (S,Tst,H)
                 const T* __restrict__ x,        // (S,Tst,Cin)
                 const T* __restrict__ cseq_g,   // model dtype (see fwd)
@@ -474,11 +476,26 @@ lstm_bwd_kernel(const T* __restrict__ dout,     // (S,H) or (S,Tst,H)
       return cseq_g + (lay_base + t) * (S_pad * RNN_H)
              + (long)blockIdx.x * (ST * RNN_H) + (wv * MT) * 64 * 4;
     };
+    // dh hand-off is FRAG-NATIVE: the GEMM2 writer and the pointwise reader
+    // use the identical lane mapping (row 16m+4lgrp+r, col hch), so each
+    // lane stores/loads its 4 values as one packed 8-byte word.
+    auto dh_at = [&](int t) {
+      return dh_g + (long)t * (S_pad * RNN_H)
+             + (long)blockIdx.x * (ST * RNN_H) + (wv * MT) * 64 * 4;
+    };
     frag gcur[MT][2];            // gates for step t: [m][{i|f, g|o}]
     #pragma unroll
     for (int m = 0; m < MT; ++m) {
       gcur[m][0] = *(((const frag*)(g_at(Tst - 1) + (m * 64 + lane) * 16)) + 0);
       gcur[m][1] = *(((const frag*)(g_at(Tst - 1) + (m * 64 + lane) * 16)) + 1);
+    }
+    // upstream dh (layer above) for step t, prefetched one stage ahead so
+    // the global loads never sit at the top of the pointwise critical path
+    ulong1 dhup[MT];
+    if (layer < L - 1) {
+      #pragma unroll
+      for (int m = 0; m < MT; ++m)
+        dhup[m] = *(const ulong1*)(dh_at(Tst - 1) + (m * 64 + lane) * 4);
     }
     // layer boundary: dA-buffer parity reuse needs the readers done, and
     // this block's dh_g stores must be visible to its next-layer loads
@@ -517,7 +534,7 @@ lstm_bwd_kernel(const T* __restrict__ dout,     // (S,H) or (S,Tst,H)
           const int row = 16 * m + 4 * lgrp + r;
           float dh = dh_rec[m][r];
           if (layer < L - 1) {
-            dh += toF<T>(dh_g[((long)t * S_pad + s0 + row) * RNN_H + hch]);
+            dh += elemF(((elem*)&dhup[m])[r]);
           } else if (ret_seq) {
             if (s0 + row < S) dh += toF<T>(dout[((long)(s0 + row) * Tst + t) * RNN_H + hch]);
           } else if (t == Tst - 1) {
@@ -561,13 +578,19 @@ lstm_bwd_kernel(const T* __restrict__ dout,     // (S,H) or (S,Tst,H)
       }
       __syncthreads();   // dA visible to all waves
 
-      // prefetch next step's gates (overlaps the dA stream + GEMMs below)
+      // prefetch next step's gates + upstream dh (overlap the GEMMs below)
       frag gnext[MT][2];
+      ulong1 dhnext[MT];
       if (t > 0) {
         #pragma unroll
         for (int m = 0; m < MT; ++m) {
           gnext[m][0] = *(((const frag*)(g_at(t - 1) + (m * 64 + lane) * 16)) + 0);
           gnext[m][1] = *(((const frag*)(g_at(t - 1) + (m * 64 + lane) * 16)) + 1);
+        }
+        if (layer < L - 1) {
+          #pragma unroll
+          for (int m = 0; m < MT; ++m)
+            dhnext[m] = *(const ulong1*)(dh_at(t - 1) + (m * 64 + lane) * 4);
         }
       }
 
@@ -609,12 +632,12 @@ lstm_bwd_kernel(const T* __restrict__ dout,     // (S,H) or (S,Tst,H)
         }
         if (layer > 0) {
           #pragma unroll
-          for (int m = 0; m < MT; ++m)
+          for (int m = 0; m < MT; ++m) {
+            T v4[4];
             #pragma unroll
-            for (int r = 0; r < 4; ++r) {
-              const int row = 16 * m + 4 * lgrp + r;
-              dh_g[((long)t * S_pad + s0 + row) * RNN_H + hch] = fromF<T>(acc[m][r]);
-            }
+            for (int r = 0; r < 4; ++r) v4[r] = fromF<T>(acc[m][r]);
+            *(ulong1*)(dh_at(t) + (m * 64 + lane) * 4) = *(ulong1*)v4;
+          }
         } else {
           #pragma unroll
           for (int m = 0; m < MT; ++m)
@@ -660,6 +683,7 @@ lstm_bwd_kernel(const T* __restrict__ dout,     // (S,H) or (S,Tst,H)
         for (int m = 0; m < MT; ++m) {
           gcur[m][0] = gnext[m][0];
           gcur[m][1] = gnext[m][1];
+          if (layer < L - 1) dhup[m] = dhnext[m];
         }
       }
     }  // t loop
