@@ -453,6 +453,10 @@ lstm_bwd_kernel(const T* __restrict__ dout,     // (S,H) or (S,Tst,H)
   char* dA_buf0 = lds;                        // [ST][512B] swizzled
   char* dA_buf1 = dA_buf0 + ST * 512;
   float* red = (float*)(dA_buf1 + ST * 512);  // [4][ST] cross-wave scratch
+  // lane-PRIVATE gates staging (each lane writes exactly the 2*MT fragments
+  // it alone reads next stage -> no cross-wave visibility, no barrier, and
+  // the prefetched tile costs LDS instead of 32 live VGPRs)
+  char* gstage = (char*)(red + 4 * ST);       // [256][MT*32B]
 
   for (int layer = L - 1; layer >= 0; --layer) {
     const bool l0cin1 = CIN1 && layer == 0;
@@ -483,11 +487,13 @@ lstm_bwd_kernel(const T* __restrict__ dout,     // (S,H) or (S,Tst,H)
       return dh_g + (long)t * (S_pad * RNN_H)
              + (long)blockIdx.x * (ST * RNN_H) + (wv * MT) * 64 * 4;
     };
-    frag gcur[MT][2];            // gates for step t: [m][{i|f, g|o}]
+    char* gmine = gstage + threadIdx.x * (MT * 32);
     #pragma unroll
     for (int m = 0; m < MT; ++m) {
-      gcur[m][0] = *(((const frag*)(g_at(Tst - 1) + (m * 64 + lane) * 16)) + 0);
-      gcur[m][1] = *(((const frag*)(g_at(Tst - 1) + (m * 64 + lane) * 16)) + 1);
+      *(frag*)&gmine[m * 32 + 0] =
+          *(((const frag*)(g_at(Tst - 1) + (m * 64 + lane) * 16)) + 0);
+      *(frag*)&gmine[m * 32 + 16] =
+          *(((const frag*)(g_at(Tst - 1) + (m * 64 + lane) * 16)) + 1);
     }
     // upstream dh (layer above) for step t, prefetched one stage ahead so
     // the global loads never sit at the top of the pointwise critical path
@@ -496,6 +502,16 @@ lstm_bwd_kernel(const T* __restrict__ dout,     // (S,H) or (S,Tst,H)
       #pragma unroll
       for (int m = 0; m < MT; ++m)
         dhup[m] = *(const ulong1*)(dh_at(Tst - 1) + (m * 64 + lane) * 4);
+    }
+    // c rotation: stage t consumes c_t and c_{t-1}; the next stage reuses
+    // c_{t-1} as ITS c_t, so only ONE new (prefetched) c load per stage.
+    ulong1 cc_t[MT], cc_p[MT];
+    #pragma unroll
+    for (int m = 0; m < MT; ++m) {
+      cc_t[m] = *(const ulong1*)(c_at(Tst - 1) + (m * 64 + lane) * 4);
+      cc_p[m] = ulong1{0};
+      if (!GRU && Tst >= 2)
+        cc_p[m] = *(const ulong1*)(c_at(Tst - 2) + (m * 64 + lane) * 4);
     }
     // layer boundary: dA-buffer parity reuse needs the readers done, and
     // this block's dh_g stores must be visible to its next-layer loads
@@ -515,19 +531,13 @@ lstm_bwd_kernel(const T* __restrict__ dout,     // (S,H) or (S,Tst,H)
 
       #pragma unroll
       for (int m = 0; m < MT; ++m) {
-        frag gf0 = gcur[m][0];                                       // i|f
-        frag gf1 = gcur[m][1];                                       // g|o
+        frag gf0 = *(const frag*)&gmine[m * 32 + 0];                 // i|f
+        frag gf1 = *(const frag*)&gmine[m * 32 + 16];                // g|o
         f32x4 ct, cpv;
-        {
-          frag ctv = {}, cpvv = {};
-          *(ulong1*)&ctv = *(const ulong1*)(c_at(t) + (m * 64 + lane) * 4);
-          if (t > 0)
-            *(ulong1*)&cpvv = *(const ulong1*)(c_at(t - 1) + (m * 64 + lane) * 4);
-          #pragma unroll
-          for (int r = 0; r < 4; ++r) {
-            ct[r] = elemF(((typename Frag8<T>::elem*)&ctv)[r]);
-            cpv[r] = (t > 0) ? elemF(((typename Frag8<T>::elem*)&cpvv)[r]) : 0.f;
-          }
+        #pragma unroll
+        for (int r = 0; r < 4; ++r) {
+          ct[r] = elemF(((elem*)&cc_t[m])[r]);
+          cpv[r] = (t > 0) ? elemF(((elem*)&cc_p[m])[r]) : 0.f;
         }
         #pragma unroll
         for (int r = 0; r < 4; ++r) {
@@ -578,14 +588,25 @@ lstm_bwd_kernel(const T* __restrict__ dout,     // (S,H) or (S,Tst,H)
       }
       __syncthreads();   // dA visible to all waves
 
-      // prefetch next step's gates + upstream dh (overlap the GEMMs below)
-      frag gnext[MT][2];
-      ulong1 dhnext[MT];
+      // prefetch next stage's saves (overlap the GEMMs below): gates go
+      // straight to the lane-private LDS slot (own-lane WAR only — this
+      // stage's reads already happened), c_{t-2} and dh to packed registers
+      ulong1 dhnext[MT], cnext[MT];
       if (t > 0) {
         #pragma unroll
         for (int m = 0; m < MT; ++m) {
-          gnext[m][0] = *(((const frag*)(g_at(t - 1) + (m * 64 + lane) * 16)) + 0);
-          gnext[m][1] = *(((const frag*)(g_at(t - 1) + (m * 64 + lane) * 16)) + 1);
+          *(frag*)&gmine[m * 32 + 0] =
+              *(((const frag*)(g_at(t - 1) + (m * 64 + lane) * 16)) + 0);
+          *(frag*)&gmine[m * 32 + 16] =
+              *(((const frag*)(g_at(t - 1) + (m * 64 + lane) * 16)) + 1);
+        }
+        // LSTM rotates c_{t-1} into c_t, so only c_{t-2} is new; GRU's
+        // save is h_{t-1} (no reuse), so its next stage needs c_at(t-1)
+        if (GRU || t >= 2) {
+          #pragma unroll
+          for (int m = 0; m < MT; ++m)
+            cnext[m] = *(const ulong1*)(c_at(GRU ? t - 1 : t - 2)
+                                        + (m * 64 + lane) * 4);
         }
         if (layer < L - 1) {
           #pragma unroll
@@ -681,8 +702,8 @@ lstm_bwd_kernel(const T* __restrict__ dout,     // (S,H) or (S,Tst,H)
       if (t > 0) {
         #pragma unroll
         for (int m = 0; m < MT; ++m) {
-          gcur[m][0] = gnext[m][0];
-          gcur[m][1] = gnext[m][1];
+          cc_t[m] = GRU ? cnext[m] : cc_p[m];
+          if (!GRU) cc_p[m] = (t >= 2) ? cnext[m] : ulong1{0};
           if (layer < L - 1) dhup[m] = dhnext[m];
         }
       }
@@ -697,7 +718,8 @@ void launch_bwd(hipStream_t stream, const void* dout, const void* x,
                 int cin, int ret_seq, int gru) {
   constexpr int ST = SEQ_TILE;
   const int nblk = (S + ST - 1) / ST;
-  const size_t lds_bytes = 2 * ST * 512 + 4 * ST * sizeof(float);
+  const size_t lds_bytes = 2 * ST * 512 + 4 * ST * sizeof(float)
+                           + 256 * (ST / 16) * 32;   // + lane-private gstage
   auto go = [&](auto kern) {
     hipLaunchKernelGGL(kern, dim3(nblk), dim3(256), lds_bytes, stream,
                        (const T*)dout, (const T*)x, (const T*)cseq_g,
