@@ -588,17 +588,19 @@ lstm_bwd_kernel(const T* __restrict__ dout,     // (S,H) or (S,Tst,H)
       }
       __syncthreads();   // dA visible to all waves
 
-      // prefetch next stage's saves (overlap the GEMMs below): gates go
-      // straight to the lane-private LDS slot (own-lane WAR only — this
-      // stage's reads already happened), c_{t-2} and dh to packed registers
+      // prefetch next stage's saves (overlap the GEMMs below). The gate
+      // loads issue NOW but their LDS stage-in happens at the END of the
+      // stage: a ds_write forces a vector-memory wait on its operand, and
+      // doing that here would drain the dA stream stores of the previous
+      // stage too (full HBM store latency exposed every stage — the 70%
+      // wave-park of round 1). c_{t-2} and dh go to packed registers.
       ulong1 dhnext[MT], cnext[MT];
+      frag gld[MT][2];
       if (t > 0) {
         #pragma unroll
         for (int m = 0; m < MT; ++m) {
-          *(frag*)&gmine[m * 32 + 0] =
-              *(((const frag*)(g_at(t - 1) + (m * 64 + lane) * 16)) + 0);
-          *(frag*)&gmine[m * 32 + 16] =
-              *(((const frag*)(g_at(t - 1) + (m * 64 + lane) * 16)) + 1);
+          gld[m][0] = *(((const frag*)(g_at(t - 1) + (m * 64 + lane) * 16)) + 0);
+          gld[m][1] = *(((const frag*)(g_at(t - 1) + (m * 64 + lane) * 16)) + 1);
         }
         // LSTM rotates c_{t-1} into c_t, so only c_{t-2} is new; GRU's
         // save is h_{t-1} (no reuse), so its next stage needs c_at(t-1)
@@ -622,7 +624,7 @@ lstm_bwd_kernel(const T* __restrict__ dout,     // (S,H) or (S,Tst,H)
         f32x4 acc[MT];
         #pragma unroll
         for (int m = 0; m < MT; ++m) acc[m] = f32x4{0.f, 0.f, 0.f, 0.f};
-        #pragma unroll 2
+        #pragma unroll 4
         for (int kk = 0; kk < 8; ++kk) {
           frag b = *(const frag*)&WhhT[hch * (4 * RNN_H) + kk * 32 + lgrp * 8];
           #pragma unroll
@@ -642,7 +644,7 @@ lstm_bwd_kernel(const T* __restrict__ dout,     // (S,H) or (S,Tst,H)
         f32x4 acc[MT];
         #pragma unroll
         for (int m = 0; m < MT; ++m) acc[m] = f32x4{0.f, 0.f, 0.f, 0.f};
-        #pragma unroll 2
+        #pragma unroll 4
         for (int kk = 0; kk < 8; ++kk) {
           frag b = *(const frag*)&WihT[hch * (4 * RNN_H) + kk * 32 + lgrp * 8];
           #pragma unroll
@@ -702,6 +704,8 @@ lstm_bwd_kernel(const T* __restrict__ dout,     // (S,H) or (S,Tst,H)
       if (t > 0) {
         #pragma unroll
         for (int m = 0; m < MT; ++m) {
+          *(frag*)&gmine[m * 32 + 0] = gld[m][0];   // loads landed long ago
+          *(frag*)&gmine[m * 32 + 16] = gld[m][1];
           cc_t[m] = GRU ? cnext[m] : cc_p[m];
           if (!GRU) cc_p[m] = (t >= 2) ? cnext[m] : ulong1{0};
           if (layer < L - 1) dhup[m] = dhnext[m];

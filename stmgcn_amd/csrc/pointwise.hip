@@ -269,12 +269,26 @@ __global__ void head_wgrad_kernel(const T* __restrict__ dy,
   const long chunk = (BN + gridDim.x - 1) / gridDim.x;
   const long r0 = (long)blockIdx.x * chunk;
   const long r1 = (r0 + chunk < BN) ? r0 + chunk : BN;
-  float acc = 0.f, accb = 0.f;
-  for (long r = r0 + wv; r < r1; r += 4) {
-    const float d = toF<T>(dy[r]);        // wave-uniform scalar load
+  // two independent accumulator chains keep 2+ row loads in flight
+  float acc = 0.f, acc2 = 0.f, accb = 0.f, accb2 = 0.f;
+  long r = r0 + wv;
+  for (; r + 4 < r1; r += 8) {
+    const float d = toF<T>(dy[r]);        // wave-uniform scalar loads
+    const float d2 = toF<T>(dy[r + 4]);
+    if (lane < G) {
+      acc += d * toF<T>(fsum[r * G + lane]);
+      acc2 += d2 * toF<T>(fsum[(r + 4) * G + lane]);
+    }
+    accb += d;
+    accb2 += d2;
+  }
+  for (; r < r1; r += 4) {
+    const float d = toF<T>(dy[r]);
     if (lane < G) acc += d * toF<T>(fsum[r * G + lane]);
     accb += d;
   }
+  acc += acc2;
+  accb += accb2;
   __shared__ float red[4][64];
   __shared__ float redb[4];
   red[wv][lane] = acc;
@@ -493,8 +507,8 @@ void stmgcn_head_bwd(void* stream, int dtype, const void* dy, const void* w,
 void stmgcn_head_wgrad(void* stream, int dtype, const void* dy,
                        const void* fsum, float* dw, float* db, int G,
                        long BN) {
-  long nblk = (BN + 1023) / 1024;
-  if (nblk > 256) nblk = 256;
+  long nblk = (BN + 255) / 256;   // fill the chip: 128+ blocks at bench size
+  if (nblk > 512) nblk = 512;
   if (nblk < 1) nblk = 1;
   dim3 grid((unsigned)nblk);
   switch (dtype) {
