@@ -189,6 +189,28 @@ lstm_fwd_kernel(const T* __restrict__ x,    // (S, Tst, C_in)
       #pragma unroll
       for (int r = 0; r < 4; ++r) c_state[m][r] = 0.f;
 
+    // input-term fragments, prefetched one stage ahead (single-buffered:
+    // reloaded right after their MFMAs consume them, so the global loads
+    // get a full stage of slack instead of stalling inside the MFMA loop)
+    frag xfr[MT][2];
+    auto load_input = [&](int t) {
+      #pragma unroll
+      for (int m = 0; m < MT; ++m) {
+        const int row = 16 * m + l16;
+        #pragma unroll
+        for (int kk = 0; kk < 2; ++kk) {
+          if (layer == 0) {            // dense x (S, Tst, 64): guard the tail
+            xfr[m][kk] = frag{};
+            if (s0 + row < S)
+              xfr[m][kk] = *(const frag*)&x[((long)(s0 + row) * Tst + t) * RNN_H + kk * 32 + lgrp * 8];
+          } else {
+            xfr[m][kk] = *(const frag*)&hin[((long)t * S_pad + s0 + row) * RNN_H + kk * 32 + lgrp * 8];
+          }
+        }
+      }
+    };
+    if (!CIN1 || layer > 0) load_input(0);
+
     for (int t = 0; t < Tst; ++t) {
       f32x4 acc[MT][4];  // [m][q]
       #pragma unroll
@@ -212,27 +234,17 @@ lstm_fwd_kernel(const T* __restrict__ x,    // (S, Tst, C_in)
         }
       }
       // input term: previous layer's h from GLOBAL (L2-hot: this block wrote
-      // the same tile last layer); dense layer 0 reads x directly
+      // the same tile last layer); dense layer 0 reads x directly. The
+      // fragments were prefetched last stage; reload for t+1 right away.
       if (!CIN1 || layer > 0) {
         #pragma unroll
-        for (int m = 0; m < MT; ++m) {
-          const int row = 16 * m + l16;
+        for (int m = 0; m < MT; ++m)
           #pragma unroll
-          for (int kk = 0; kk < 2; ++kk) {
-            frag a;
-            if (layer == 0) {          // dense x (S, Tst, 64): guard the tail
-              if (s0 + row < S)
-                a = *(const frag*)&x[((long)(s0 + row) * Tst + t) * RNN_H + kk * 32 + lgrp * 8];
-              else
-                a = frag{};
-            } else {
-              a = *(const frag*)&hin[((long)t * S_pad + s0 + row) * RNN_H + kk * 32 + lgrp * 8];
-            }
+          for (int kk = 0; kk < 2; ++kk)
             #pragma unroll
             for (int q = 0; q < 4; ++q)
-              acc[m][q] = mfma16x16x32(a, bxfrag[q][kk], acc[m][q]);
-          }
-        }
+              acc[m][q] = mfma16x16x32(xfr[m][kk], bxfrag[q][kk], acc[m][q]);
+        if (t + 1 < Tst) load_input(t + 1);
       }
 
       // pointwise cell update in fragment layout (+ scalar-input term)
