@@ -236,6 +236,8 @@ def main():
                 "parallelism": f"dp{n_gpus}",
                 "hipgraph": graph_ok,
                 "impl": impl,
+                "max_mem_gb": round(torch.cuda.max_memory_allocated() / 2**30, 1)
+                              if use_gpu else 0.0,
             },
         }
         print(json.dumps(rec))

@@ -1,5 +1,7 @@
 """GPU numerics tests: every HIP kernel vs the fp32 pure-PyTorch oracle
 (SURVEY §4 item 1). All marked @pytest.mark.gpu."""
+import os
+
 import numpy as np
 import pytest
 import torch
@@ -706,3 +708,23 @@ def test_cheb_gconv_fused_no_bias_no_act():
         rel = ((got.float().cpu() - want).abs().max() /
                (want.abs().max() + 1e-6)).item()
         assert rel < 0.08
+
+
+def test_perf_regression_gate():
+    """a short bench run must stay above a samples/s floor
+    on the pure-hip path — catches silent fallbacks (impl would report
+    hip+fallback) and performance cliffs at driver time. The floor (9000)
+    is ~20% under the measured ~11k steady state on a fresh MI355X."""
+    import json
+    import subprocess
+    import sys
+    repo = os.path.dirname(os.path.dirname(os.path.abspath(__file__)))
+    r = subprocess.run(
+        [sys.executable, os.path.join(repo, "bench.py"),
+         "--steps", "10", "--warmup", "5"],
+        capture_output=True, text=True, timeout=600, cwd=repo)
+    assert r.returncode == 0, r.stderr[-2000:]
+    rec = json.loads([l for l in r.stdout.splitlines() if l.startswith("{")][0])
+    assert rec["config"]["impl"] == "hip", rec["config"]
+    assert rec["config"]["hipgraph"] is True, "whole-step capture regressed"
+    assert rec["value"] > 9000, f"perf cliff: {rec['value']:.0f} samples/s"
