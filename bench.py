@@ -60,6 +60,11 @@ def main():
 
     if args.impl:
         os.environ["STMGCN_IMPL"] = args.impl
+    # Stock-floor measurability: MIOpen's RNN find can run for >10 min at the
+    # deep/large shapes (MIOPEN_FIND_MODE only governs convolutions), so the
+    # floor can optionally be measured on torch's decomposed native RNN path.
+    if os.environ.get("STMGCN_TORCH_RNN_NATIVE", "0") == "1":
+        torch.backends.cudnn.enabled = False
     cfg = PRESETS[args.preset]
     if args.batch_size:
         cfg = cfg.replace(batch_size=args.batch_size)
