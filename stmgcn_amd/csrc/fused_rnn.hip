@@ -2,12 +2,14 @@
 //
 // Replaces the reference's nn.LSTM over the flattened (B*N, T, C) sequence
 // batch (STMGCN.py:21-22,47-50). One kernel runs ALL layers and ALL
-// timesteps: each 256-thread workgroup owns a 64-sequence tile, stages the
-// evolving per-timestep hidden sequence in LDS (XOR-swizzled for
-// conflict-free ds_read_b128 MFMA fragment reads), computes the recurrent +
-// input projections with v_mfma_f32_16x16x32_bf16 matrix cores (fp32
-// accumulate), and applies the gate nonlinearities in the MFMA accumulator
-// fragment layout, so hidden state never leaves the CU between timesteps.
+// timesteps: each 256-thread workgroup owns a 32-sequence tile, keeps the
+// LIVE hidden state (t-1/t ping/pong) in XOR-swizzled LDS for conflict-free
+// ds_read_b128 MFMA fragment reads, computes the recurrent + input
+// projections with v_mfma_f32_16x16x32_bf16 matrix cores (fp32 accumulate),
+// and applies the gate nonlinearities in the MFMA accumulator fragment
+// layout. The cross-LAYER sequence hand-off rides hseq_g in global memory
+// (mandatory for the wgrad kernel anyway), with the input fragments
+// prefetched one stage ahead.
 //
 // Geometry per workgroup (H = 64 fixed):
 //   4 waves; wave w owns gate-channel slice [16w, 16w+16) of each of the
