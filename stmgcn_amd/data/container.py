@@ -49,7 +49,9 @@ class DataInput:
         self._max: Optional[float] = None
 
     def load_data(self) -> Dict[str, np.ndarray]:
+        print("Loading data...")                 # reference Data_Container.py:15
         npz = np.load(self.data_dir)
+        print("Available keys:", list(npz.keys()))
         dataset: Dict[str, np.ndarray] = {}
         dataset["taxi"] = self.minmax_normalize(npz["taxi"]) if self.norm_opt else npz["taxi"]
         for k in range(self.M_sta):
@@ -66,6 +68,7 @@ class DataInput:
 
     def minmax_normalize(self, x: np.ndarray) -> np.ndarray:
         self._max, self._min = float(x.max()), float(x.min())
+        print("min:", self._min, "max:", self._max)  # Data_Container.py:33
         x = (x - self._min) / (self._max - self._min)
         return 2.0 * x - 1.0
 
@@ -73,6 +76,16 @@ class DataInput:
         if self._max is None:
             raise RuntimeError("minmax_denormalize called before normalize")
         return (self._max - self._min) * ((x + 1.0) / 2.0) + self._min
+
+    # std pair: dead code in the reference (Data_Container.py:43-51 — nothing
+    # calls it) but part of the DataInput API surface; reproduced for parity.
+    def std_normalize(self, x: np.ndarray) -> np.ndarray:
+        self._mean, self._std = float(x.mean()), float(x.std())
+        print("mean:", round(self._mean, 4), "std:", round(self._std, 4))
+        return (x - self._mean) / self._std
+
+    def std_denormalize(self, x: np.ndarray) -> np.ndarray:
+        return x * self._std + self._mean
 
 
 def sliding_windows(data: np.ndarray, serial_len: int, daily_len: int,

@@ -134,3 +134,28 @@ def test_device_loader_drop_last_when_sharded():
     dl1 = DeviceLoader(x, y, batch_size=b)
     sizes = [xb.shape[0] for xb, _ in dl1]
     assert sum(sizes) == n and sizes[-1] == n % b
+
+
+def test_std_normalize_pair_and_load_prints(capsys, tmp_path):
+    """Reference API-surface parity: the (dead in the reference) std
+    normalize/denormalize pair round-trips, and load_data reproduces the
+    reference's prints (Data_Container.py:15-17,33,43-51)."""
+    import numpy as np
+    from stmgcn_amd.data import DataInput
+    rng = np.random.default_rng(0)
+    x = rng.normal(5.0, 2.0, size=(50, 4, 1))
+    di = DataInput(M_adj=1, data_dir="", norm_opt=True)
+    z = di.std_normalize(x)
+    assert abs(z.mean()) < 1e-9 and abs(z.std() - 1.0) < 1e-9
+    np.testing.assert_allclose(di.std_denormalize(z), x, rtol=1e-12)
+    out = capsys.readouterr().out
+    assert out.startswith("mean:") and "std:" in out
+
+    npz = tmp_path / "d.npz"
+    np.savez(npz, taxi=x, neighbor_adj=np.eye(4))
+    di2 = DataInput(M_adj=1, data_dir=str(npz), norm_opt=True)
+    di2.load_data()
+    out = capsys.readouterr().out
+    assert "Loading data..." in out
+    assert "Available keys:" in out
+    assert "min:" in out and "max:" in out
