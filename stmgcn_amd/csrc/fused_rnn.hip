@@ -729,6 +729,289 @@ lstm_bwd_kernel(const T* __restrict__ dout,     // (S,H) or (S,Tst,H)
   }  // layer loop
 }
 
+
+// ===========================================================================
+// WAVE-PRIVATE BPTT backward (round-2 experiment, STMGCN_BWD_WAVE=1).
+//
+// The classic lstm_bwd_kernel couples its 4 waves every stage: the
+// pointwise phase writes dA columns that OTHER waves' GEMMs consume, so a
+// __syncthreads sits on the serial BPTT critical path 24 times per block —
+// measured 71% wave-park at occupancy 2 (profiles/r02_pmc_sq_waits.md).
+//
+// This variant gives each WAVE 16 sequences end-to-end: pointwise (in the
+// MFMA D-fragment layout, which is ALSO the forward save layout, so gates/
+// cell bundles arrive as coalesced 16 B fragments), a wave-private LDS
+// transpose of dA into A-fragment layout, and both GEMMs over the full
+// k = 4H and n = H ranges per wave. Nothing crosses waves: there are ZERO
+// barriers in the kernel, the cross-layer dh hand-off is packed so the
+// writing lane is the reading lane (plain vmcnt ordering suffices), and a
+// 64-row block doubles the rows in flight per CU at the same occupancy.
+//
+// Geometry: 256 threads = 4 waves; wave w owns rows [64*blockIdx + 16w ..
+// +16). D-layout per lane: rows 16w + 4*lgrp + r (r 0..3), channels
+// 16*nt + l16 (nt 0..3). Forward saved its bundles from 32-row blocks with
+// channel-sliced waves; the (block, m, wave, lane) coordinates of this
+// wave's rows are bxf = row0/32, mf = (row0/16)&1, wv_f = nt, lane_f = lane
+// — the SAME lane index, so every gates/cseq access is a direct fragment
+// load of the forward's save.
+template <typename T, bool CIN1, bool GRU>
+__global__ void __launch_bounds__(256, 2)
+lstm_bwd_wave_kernel(const T* __restrict__ dout,    // (S,H) or (S,Tst,H)
+                     const T* __restrict__ cseq_g,  // fwd save (see fwd)
+                     const T* __restrict__ gates_g, // fwd save
+                     RnnPtrs w,                     // w_ih/w_hh TRANSPOSED (C|H, 4H)
+                     T* __restrict__ dx,            // (S,Tst,Cin)
+                     T* __restrict__ dA_g,          // (L,Tst,S_pad,4H) natural
+                     T* __restrict__ dh_g,          // (Tst*S_pad*H) packed scratch
+                     int S, int S_pad, int Tst, int L, int ret_seq) {
+  using frag = typename Frag8<T>::type;
+  using elem = typename Frag8<T>::elem;
+  extern __shared__ char lds[];
+  const int wv = threadIdx.x >> 6;
+  const int lane = threadIdx.x & 63;
+  const int l16 = lane & 15;
+  const int lgrp = lane >> 4;
+  const int row0 = blockIdx.x * 64 + 16 * wv;     // wave's first row
+  if (row0 >= S_pad) return;                       // whole-wave tail guard
+
+  char* dAw = lds + wv * 8192;                     // wave-private dA (16x512B)
+  char* gmine = lds + 32768 + threadIdx.x * 128;   // lane-private gates stage
+
+  const int bxf = row0 >> 5;                       // fwd 32-row block
+  const int mf = (row0 >> 4) & 1;                  // fwd m-tile
+
+  for (int layer = L - 1; layer >= 0; --layer) {
+    const bool l0cin1 = CIN1 && layer == 0;
+    const T* WhhT = (const T*)w.w_hh[layer];       // (H, 4H)
+    const T* WihT = (const T*)w.w_ih[layer];       // (Cin_l, 4H)
+    const long lay_base = (long)layer * Tst;
+    // fwd-save bundle addresses for this lane (16 gate T / 4 cell T per nt)
+    auto g_at = [&](int t, int nt) {
+      return gates_g + (lay_base + t) * ((long)S_pad * 4 * RNN_H)
+             + (long)bxf * (32 * 4 * RNN_H) + nt * (2 * 64 * 16)
+             + (mf * 64 + lane) * 16;
+    };
+    auto c_at = [&](int t, int nt) {
+      return cseq_g + (lay_base + t) * ((long)S_pad * RNN_H)
+             + (long)bxf * (32 * RNN_H) + nt * (2 * 64 * 4)
+             + (mf * 64 + lane) * 4;
+    };
+    // packed lane-private dh hand-off (writer lane == reader lane)
+    auto dh_at = [&](int t) {
+      return dh_g + (long)t * ((long)S_pad * RNN_H)
+             + (long)blockIdx.x * (64 * RNN_H) + wv * (16 * RNN_H) + lane * 16;
+    };
+
+    float dh_rec[4][4], dc[4][4];                  // [nt][r], D-layout
+    #pragma unroll
+    for (int nt = 0; nt < 4; ++nt)
+      #pragma unroll
+      for (int r = 0; r < 4; ++r) { dh_rec[nt][r] = 0.f; dc[nt][r] = 0.f; }
+
+    float wihv[4][4];                              // [q][nt] this lane's W_ih column
+    if (l0cin1)
+      #pragma unroll
+      for (int q = 0; q < 4; ++q)
+        #pragma unroll
+        for (int nt = 0; nt < 4; ++nt)
+          wihv[q][nt] = toF<T>(WihT[q * 64 + 16 * nt + l16]);
+
+    // ---- stage-in for t = Tst-1: gates -> lane LDS, c/dh -> registers ----
+    #pragma unroll
+    for (int nt = 0; nt < 4; ++nt) {
+      *(frag*)&gmine[nt * 32 + 0] = *(((const frag*)g_at(Tst - 1, nt)) + 0);
+      *(frag*)&gmine[nt * 32 + 16] = *(((const frag*)g_at(Tst - 1, nt)) + 1);
+    }
+    ulong1 cc_t[4], cc_p[4], dhup[4];
+    #pragma unroll
+    for (int nt = 0; nt < 4; ++nt) {
+      cc_t[nt] = *(const ulong1*)c_at(Tst - 1, nt);
+      cc_p[nt] = ulong1{0};
+      if (!GRU && Tst >= 2) cc_p[nt] = *(const ulong1*)c_at(Tst - 2, nt);
+    }
+    if (layer < L - 1) {
+      #pragma unroll
+      for (int nt = 0; nt < 4; ++nt)
+        dhup[nt] = *(((const ulong1*)dh_at(Tst - 1)) + nt);
+    }
+
+    for (int t = Tst - 1; t >= 0; --t) {
+      // ---- pointwise in D-layout; dA transposed through wave LDS --------
+      float dxacc[4];                              // [r] (CIN1 l0 only)
+      if (l0cin1) { dxacc[0] = dxacc[1] = dxacc[2] = dxacc[3] = 0.f; }
+      #pragma unroll 1
+      for (int nt = 0; nt < 4; ++nt) {             // serialized: live-range cap
+        frag gf0 = *(const frag*)&gmine[nt * 32 + 0];   // [i r0..3 | f r0..3]
+        frag gf1 = *(const frag*)&gmine[nt * 32 + 16];  // [g r0..3 | o r0..3]
+        #pragma unroll
+        for (int r = 0; r < 4; ++r) {
+          const int row = row0 + 4 * lgrp + r;
+          float dh = dh_rec[nt][r];
+          if (layer < L - 1) {
+            dh += elemF(((elem*)&dhup[nt])[r]);
+          } else if (ret_seq) {
+            if (row < S)
+              dh += toF<T>(dout[((long)row * Tst + t) * RNN_H + 16 * nt + l16]);
+          } else if (t == Tst - 1) {
+            if (row < S)
+              dh += toF<T>(dout[(long)row * RNN_H + 16 * nt + l16]);
+          }
+          const float ctv = elemF(((elem*)&cc_t[nt])[r]);
+          float dAi, dAf, dAg, dAo;
+          if (GRU) {
+            const float r_ = elemF(gf0[r]);
+            const float z_ = elemF(gf0[4 + r]);
+            const float n_ = elemF(gf1[r]);
+            const float Bn = elemF(gf1[4 + r]);
+            const float hprev = ctv;               // GRU save = h_{t-1}
+            dh += dc[nt][r];
+            const float dz_pre = dh * (hprev - n_) * z_ * (1.f - z_);
+            const float dn_pre = dh * (1.f - z_) * (1.f - n_ * n_);
+            dAo = dn_pre * r_;                     // dBn
+            dAi = dn_pre * Bn * r_ * (1.f - r_);   // dr_pre
+            dAf = dz_pre;
+            dAg = dn_pre;
+            dc[nt][r] = dh * z_;
+          } else {
+            const float i_ = elemF(gf0[r]);
+            const float f_ = elemF(gf0[4 + r]);
+            const float g_ = elemF(gf1[r]);
+            const float o_ = elemF(gf1[4 + r]);
+            const float cpv = (t > 0) ? elemF(((elem*)&cc_p[nt])[r]) : 0.f;
+            const float tc = stm_tanh(ctv);
+            float dcv = dc[nt][r] + dh * o_ * (1.f - tc * tc);
+            dAo = dh * tc * o_ * (1.f - o_);
+            dAi = dcv * g_ * i_ * (1.f - i_);
+            dAf = dcv * cpv * f_ * (1.f - f_);
+            dAg = dcv * i_ * (1.f - g_ * g_);
+            dc[nt][r] = dcv * f_;
+          }
+          const int rl = 4 * lgrp + r;             // wave-local row
+          const int ch = 16 * nt + l16;
+          *(T*)&dAw[swzA(rl, (0 * 64 + ch) * 2)] = fromF<T>(dAi);
+          *(T*)&dAw[swzA(rl, (1 * 64 + ch) * 2)] = fromF<T>(dAf);
+          *(T*)&dAw[swzA(rl, (2 * 64 + ch) * 2)] = fromF<T>(dAg);
+          *(T*)&dAw[swzA(rl, (3 * 64 + ch) * 2)] = fromF<T>(dAo);
+          if (l0cin1)
+            dxacc[r] += dAi * wihv[0][nt] + dAf * wihv[1][nt]
+                      + dAg * wihv[2][nt] + dAo * wihv[3][nt];
+        }
+      }
+      // wave-private transpose: in-order LDS, no cross-wave traffic — the
+      // A-fragment reads below only need the implicit lgkmcnt waits (kept
+      // in LDS, re-read per GEMM: 8 extra ds_reads beat 32 live VGPRs)
+      auto a_at = [&](int kk) {
+        return *(const frag*)&dAw[swzA(l16, (kk * 32 + lgrp * 8) * 2)];
+      };
+
+      // ---- prefetch next stage's saves (land during the GEMMs) ----------
+      frag gld[4][2];
+      ulong1 cnext[4], dhnext[4];
+      if (t > 0) {
+        #pragma unroll
+        for (int nt = 0; nt < 4; ++nt) {
+          gld[nt][0] = *(((const frag*)g_at(t - 1, nt)) + 0);
+          gld[nt][1] = *(((const frag*)g_at(t - 1, nt)) + 1);
+        }
+        if (GRU || t >= 2) {
+          #pragma unroll
+          for (int nt = 0; nt < 4; ++nt)
+            cnext[nt] = *(const ulong1*)c_at(GRU ? t - 1 : t - 2, nt);
+        }
+        if (layer < L - 1) {
+          #pragma unroll
+          for (int nt = 0; nt < 4; ++nt)
+            dhnext[nt] = *(((const ulong1*)dh_at(t - 1)) + nt);
+        }
+      }
+
+      // ---- GEMM1: dh_prev = dA @ W_hh (full k=4H, n=H per wave) ---------
+      if (t > 0) {
+        #pragma unroll 2
+        for (int nt = 0; nt < 4; ++nt) {
+          f32x4 acc = f32x4{0.f, 0.f, 0.f, 0.f};
+          #pragma unroll 4
+          for (int kk = 0; kk < 8; ++kk) {
+            frag b = *(const frag*)&WhhT[(16 * nt + l16) * (4 * RNN_H)
+                                         + kk * 32 + lgrp * 8];
+            acc = mfma16x16x32(a_at(kk), b, acc);
+          }
+          #pragma unroll
+          for (int r = 0; r < 4; ++r) dh_rec[nt][r] = acc[r];
+        }
+      }
+
+      // ---- GEMM2: dx / dh hand-off ---------------------------------------
+      if (!l0cin1) {
+        #pragma unroll 2
+        for (int nt = 0; nt < 4; ++nt) {
+          f32x4 acc = f32x4{0.f, 0.f, 0.f, 0.f};
+          #pragma unroll 4
+          for (int kk = 0; kk < 8; ++kk) {
+            frag b = *(const frag*)&WihT[(16 * nt + l16) * (4 * RNN_H)
+                                         + kk * 32 + lgrp * 8];
+            acc = mfma16x16x32(a_at(kk), b, acc);
+          }
+          if (layer > 0) {                         // packed lane-private
+            T v4[4];
+            #pragma unroll
+            for (int r = 0; r < 4; ++r) v4[r] = fromF<T>(acc[r]);
+            *(((ulong1*)dh_at(t)) + nt) = *(ulong1*)v4;
+          } else {                                 // dense dx (S,Tst,64)
+            #pragma unroll
+            for (int r = 0; r < 4; ++r) {
+              const int row = row0 + 4 * lgrp + r;
+              if (row < S)
+                dx[((long)row * Tst + t) * RNN_H + 16 * nt + l16] =
+                    fromF<T>(acc[r]);
+            }
+          }
+        }
+      } else {
+        // CIN1 l0: dx[row,t] = sum_ch dxacc; reduce over the 16 l16 lanes
+        #pragma unroll
+        for (int r = 0; r < 4; ++r) {
+          float v = dxacc[r];
+          #pragma unroll
+          for (int off = 1; off < 16; off <<= 1) v += __shfl_xor(v, off, 64);
+          const int row = row0 + 4 * lgrp + r;
+          if (l16 == 0 && row < S) dx[(long)row * Tst + t] = fromF<T>(v);
+        }
+      }
+
+      // ---- dA stream to global (natural layout, from the A-fragments) ---
+      {
+        T* out_dA = dA_g + (lay_base + t) * ((long)S_pad * 4 * RNN_H)
+                    + (long)(row0 + l16) * 4 * RNN_H;
+        #pragma unroll
+        for (int kk = 0; kk < 8; ++kk)
+          *(frag*)&out_dA[kk * 32 + lgrp * 8] = a_at(kk);
+      }
+
+      // ---- stage-end: stage the prefetched gates into lane LDS, rotate --
+      if (t > 0) {
+        #pragma unroll
+        for (int nt = 0; nt < 4; ++nt) {
+          *(frag*)&gmine[nt * 32 + 0] = gld[nt][0];
+          *(frag*)&gmine[nt * 32 + 16] = gld[nt][1];
+          cc_t[nt] = GRU ? cnext[nt] : cc_p[nt];
+          if (!GRU) cc_p[nt] = (t >= 2) ? cnext[nt] : ulong1{0};
+          if (layer < L - 1) dhup[nt] = dhnext[nt];
+        }
+      }
+    }  // t loop
+  }  // layer loop
+}
+
+static int bwd_wave_mode() {
+  static int mode = -1;
+  if (mode < 0) {
+    const char* e = getenv("STMGCN_BWD_WAVE");
+    mode = e ? atoi(e) : 0;
+  }
+  return mode;
+}
+
 template <typename T>
 void launch_bwd(hipStream_t stream, const void* dout, const void* x,
                 const void* cseq_g, const void* gates_g, const RnnPtrs& w,
@@ -736,6 +1019,22 @@ void launch_bwd(hipStream_t stream, const void* dout, const void* x,
                 int cin, int ret_seq, int gru) {
   constexpr int ST = SEQ_TILE;
   const int nblk = (S + ST - 1) / ST;
+  const int S_pad = nblk * ST;
+  if (bwd_wave_mode()) {
+    const int nblk64 = (S_pad + 63) / 64;
+    const size_t ldsw = 32768 + 256 * 128;     // wave dA slots + lane gates
+    auto gow = [&](auto kern) {
+      hipLaunchKernelGGL(kern, dim3(nblk64), dim3(256), ldsw, stream,
+                         (const T*)dout, (const T*)cseq_g, (const T*)gates_g,
+                         w, (T*)dx, (T*)dA_g, (T*)dh_g, S, S_pad, Tst, L,
+                         ret_seq);
+    };
+    if (cin == 1 && !gru) gow(lstm_bwd_wave_kernel<T, true, false>);
+    else if (cin == 1 && gru) gow(lstm_bwd_wave_kernel<T, true, true>);
+    else if (!gru) gow(lstm_bwd_wave_kernel<T, false, false>);
+    else gow(lstm_bwd_wave_kernel<T, false, true>);
+    return;
+  }
   const size_t lds_bytes = 2 * ST * 512 + 4 * ST * sizeof(float)
                            + 256 * (ST / 16) * 32;   // + lane-private gstage
   auto go = [&](auto kern) {
